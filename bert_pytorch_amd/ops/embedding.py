@@ -30,9 +30,13 @@ class _FusedEmbeddingLNDropout(torch.autograd.Function):
             if (training and p > 0)
             else (0, 0)
         )
+        if word_emb.is_cuda and torch.is_autocast_enabled():
+            out_dtype = torch.get_autocast_gpu_dtype()
+        else:
+            out_dtype = word_emb.dtype
         y, z, mask, mean, rstd = ext.embedding_ln_dropout_fwd(
             ids, tt, word_emb, pos_emb, tok_emb, weight, ln_bias,
-            p if training else 0.0, eps, seed, offset,
+            p if training else 0.0, eps, seed, offset, out_dtype,
         )
         ctx.save_for_backward(ids, tt, z, mask, weight, mean, rstd)
         ctx.p = p if training else 0.0

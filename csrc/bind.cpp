@@ -1,0 +1,79 @@
+// Python bindings for the MI355X (gfx950) HIP kernels.
+#include <torch/extension.h>
+
+namespace bpa {
+
+std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor gamma,
+                                  torch::Tensor beta, double eps);
+std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x,
+                                  torch::Tensor gamma, torch::Tensor mean,
+                                  torch::Tensor rstd);
+torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias);
+std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor bias);
+std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
+    torch::Tensor x, c10::optional<torch::Tensor> bias, torch::Tensor residual,
+    torch::Tensor gamma, torch::Tensor beta, double p, double eps,
+    int64_t seed, int64_t offset);
+std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
+    torch::Tensor dy, torch::Tensor z, torch::Tensor mask, torch::Tensor gamma,
+    torch::Tensor mean, torch::Tensor rstd, double p, bool has_bias);
+std::vector<torch::Tensor> embedding_ln_dropout_fwd(
+    torch::Tensor ids, c10::optional<torch::Tensor> tt, torch::Tensor word,
+    torch::Tensor pos, c10::optional<torch::Tensor> tok, torch::Tensor gamma,
+    torch::Tensor beta, double p, double eps, int64_t seed, int64_t offset,
+    torch::ScalarType out_dtype);
+std::vector<torch::Tensor> embedding_ln_dropout_bwd(
+    torch::Tensor dy, torch::Tensor ids, c10::optional<torch::Tensor> tt,
+    torch::Tensor z, torch::Tensor mask, torch::Tensor gamma,
+    torch::Tensor mean, torch::Tensor rstd, double p, int64_t vocab,
+    int64_t max_pos, int64_t n_types);
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels,
+                                  int64_t ignore_index);
+torch::Tensor ce_bwd(torch::Tensor dloss, torch::Tensor logits,
+                     torch::Tensor labels, torch::Tensor lse,
+                     torch::Tensor count, int64_t ignore_index);
+std::vector<torch::Tensor> attention_fwd(torch::Tensor qkv,
+                                         torch::Tensor seqlens,
+                                         int64_t num_heads, double p,
+                                         int64_t seed, int64_t offset);
+torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
+                            torch::Tensor seqlens, torch::Tensor out,
+                            torch::Tensor lse, int64_t num_heads, double p,
+                            int64_t seed, int64_t offset);
+torch::Tensor multi_tensor_l2norm_sq(std::vector<torch::Tensor> tensors);
+void multi_tensor_clip_scale(std::vector<torch::Tensor> grads,
+                             torch::Tensor gnorm_sq, double max_norm);
+void fused_lamb(std::vector<torch::Tensor> params,
+                std::vector<torch::Tensor> grads,
+                std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
+                torch::Tensor gnorm_sq, double lr, double beta1, double beta2,
+                double eps, double wd, int64_t step, bool bias_correction,
+                bool grad_averaging, double max_grad_norm, bool use_ratio);
+void fused_adam(std::vector<torch::Tensor> params,
+                std::vector<torch::Tensor> grads,
+                std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
+                double lr, double beta1, double beta2, double eps, double wd,
+                int64_t step, bool bias_correction, bool adam_w);
+
+}  // namespace bpa
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "bert_pytorch_amd gfx950 HIP kernels";
+  m.def("ln_fwd", &bpa::ln_fwd, "fused LayerNorm forward");
+  m.def("ln_bwd", &bpa::ln_bwd, "fused LayerNorm backward");
+  m.def("bias_gelu_fwd", &bpa::bias_gelu_fwd);
+  m.def("bias_gelu_bwd", &bpa::bias_gelu_bwd);
+  m.def("bias_dropout_residual_ln_fwd", &bpa::bias_dropout_residual_ln_fwd);
+  m.def("bias_dropout_residual_ln_bwd", &bpa::bias_dropout_residual_ln_bwd);
+  m.def("embedding_ln_dropout_fwd", &bpa::embedding_ln_dropout_fwd);
+  m.def("embedding_ln_dropout_bwd", &bpa::embedding_ln_dropout_bwd);
+  m.def("ce_fwd", &bpa::ce_fwd);
+  m.def("ce_bwd", &bpa::ce_bwd);
+  m.def("attention_fwd", &bpa::attention_fwd);
+  m.def("attention_bwd", &bpa::attention_bwd);
+  m.def("multi_tensor_l2norm_sq", &bpa::multi_tensor_l2norm_sq);
+  m.def("multi_tensor_clip_scale", &bpa::multi_tensor_clip_scale);
+  m.def("fused_lamb", &bpa::fused_lamb);
+  m.def("fused_adam", &bpa::fused_adam);
+}

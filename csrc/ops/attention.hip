@@ -1,0 +1,635 @@
+// Fused scaled-dot-product attention (flash-style) for MI355X (gfx950).
+//
+// Replaces the reference's 5-op attention path (QK^T, +mask, softmax,
+// dropout, PV — src/modeling.py:403-429) with MFMA kernels built on
+// v_mfma_f32_16x16x32_bf16:
+//
+// Forward: grid (S/64 q-tiles, B*heads); 4 waves/block, 16 q-rows per
+// wave. K and V^T staged in LDS (+8 bf16 row pad, conflict-free b64
+// reads); online softmax with the swapped-QK^T trick (S^T = mfma(K, Q))
+// so each lane's P scores chain directly into the PV A-fragment; the
+// padding mask enters as per-sequence valid lengths; dropout by
+// Philox4x32-10 regenerated (not stored) in backward.
+//
+// Backward (flash-2 style): grid (S/64 kv-tiles, B*heads); each block
+// owns one K/V tile, loops over q-tiles; recomputes P from the saved
+// logsumexp; accumulates dK/dV in registers (written once, exclusively,
+// straight into dqkv) and dQ via fp32 atomics into a workspace packed
+// by a final kernel.
+//
+// Fragment convention (consistent A/B slot mapping):
+//   A[i][k]/B[k][j]: i|j = lane&15, k = (lane>>4)*4 + (e&3) + 16*(e>>2)
+//   C/D[i][j]:       j = lane&15,   i = (lane>>4)*4 + reg
+// Only A/B consistency matters for correctness (the contraction is
+// permutation-invariant when both sides agree); C/D is the documented
+// gfx950 layout.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "../common.h"
+
+namespace bpa {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __hip_bfloat16 bf16_t;
+
+#define MFMA16(A, B, C) __builtin_amdgcn_mfma_f32_16x16x32_bf16((A), (B), (C), 0, 0, 0)
+
+constexpr int kPad = 8;        // bf16 row pad: 72-element stride, bank-clean
+constexpr int kStride = 64 + kPad;
+
+// Build an A/B fragment from a row-major bf16 row pointer: elements at
+// k = base + g*4 + (e&3) + 16*(e>>2), loaded as two 8-byte chunks.
+__device__ __forceinline__ bf16x8 frag_row(const __bf16* row, int base,
+                                           int g) {
+  union {
+    bf16x8 v;
+    uint2 u[2];
+  } r;
+  r.u[0] = *reinterpret_cast<const uint2*>(row + base + g * 4);
+  r.u[1] = *reinterpret_cast<const uint2*>(row + base + 16 + g * 4);
+  return r.v;
+}
+
+__device__ __forceinline__ float uniform_at(const Philox& ph, uint64_t ctr,
+                                            int word) {
+  uint32_t r4[4];
+  ph(ctr, r4);
+  return u32_to_uniform(r4[word]);
+}
+
+// ---------------------------------------------------------------------------
+// forward
+// ---------------------------------------------------------------------------
+template <bool TRAIN_DROP>
+__global__ __launch_bounds__(256) void attn_fwd_kernel(
+    const __bf16* __restrict__ qkv, const int* __restrict__ seqlens,
+    __bf16* __restrict__ out, float* __restrict__ lse_out, int B, int S,
+    int NH, float p, float scale, uint64_t seed, uint64_t offset) {
+  const int bh = blockIdx.y;
+  const int b = bh / NH, h = bh % NH;
+  const int q0 = blockIdx.x * 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int g = (lane >> 4), li = lane & 15;
+  const int H = NH * 64;
+  const int rs3 = 3 * H;  // qkv row stride
+  const __bf16* qbase = qkv + static_cast<int64_t>(b) * S * rs3 + h * 64;
+  const __bf16* kbase = qbase + H;
+  const __bf16* vbase = qbase + 2 * H;
+  const int slen = seqlens[b];
+  const float inv_keep = TRAIN_DROP ? 1.f / (1.f - p) : 1.f;
+  Philox philox(seed);
+  const int s2 = S >> 1;
+  const uint64_t drop_base =
+      offset + static_cast<uint64_t>(bh) * s2 * s2;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* K_lds = reinterpret_cast<__bf16*>(smem);             // [64][72]
+  __bf16* Vt_lds = K_lds + 64 * kStride;                       // [64][72]
+  float* alpha_lds = reinterpret_cast<float*>(Vt_lds + 64 * kStride);  // [4][16]
+  float* stat_lds = alpha_lds + 4 * 16;                        // [4][16]
+
+  // Q fragments for this wave's 16 rows (row = q0 + wave*16 + li), held
+  // in registers for the whole K/V sweep.
+  const int q_row = q0 + wave * 16 + li;
+  const int q_ld = min(q_row, S - 1);
+  bf16x8 qfrag[2];
+#pragma unroll
+  for (int c = 0; c < 2; ++c)
+    qfrag[c] = frag_row(qbase + static_cast<int64_t>(q_ld) * rs3, 32 * c, g);
+
+  float m_run = -1e30f, l_run = 0.f;
+  f32x4 acc_o[4] = {};
+
+  const int n_kv = (S + 63) / 64;
+  for (int kt = 0; kt < n_kv; ++kt) {
+    const int k0 = kt * 64;
+    __syncthreads();
+    // stage K natural + V transposed; 4 threads per row, 16 cols each
+    {
+      const int row = tid >> 2, colc = (tid & 3) * 16;
+      const int krow = k0 + row;
+      if (krow < S) {
+        const uint4* src =
+            reinterpret_cast<const uint4*>(kbase + static_cast<int64_t>(krow) * rs3 + colc);
+        *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc]) = src[0];
+        *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc + 8]) = src[1];
+        __bf16 vv[16];
+        const uint4* vsrc =
+            reinterpret_cast<const uint4*>(vbase + static_cast<int64_t>(krow) * rs3 + colc);
+        *reinterpret_cast<uint4*>(&vv[0]) = vsrc[0];
+        *reinterpret_cast<uint4*>(&vv[8]) = vsrc[1];
+#pragma unroll
+        for (int j = 0; j < 16; ++j) Vt_lds[(colc + j) * kStride + row] = vv[j];
+      } else {
+        uint4 zero{0, 0, 0, 0};
+        *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc]) = zero;
+        *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc + 8]) = zero;
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          Vt_lds[(colc + j) * kStride + row] = __bf16(0.f);
+      }
+    }
+    __syncthreads();
+
+    // S^T tiles: A = K (16 keys), B = Q^T; C[key_local][q]: q = li
+    f32x4 s_acc[4];
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      const __bf16* krow = &K_lds[(t * 16 + li) * kStride];
+      f32x4 acc = {};
+      acc = MFMA16(frag_row(krow, 0, g), qfrag[0], acc);
+      acc = MFMA16(frag_row(krow, 32, g), qfrag[1], acc);
+      s_acc[t] = acc;
+    }
+
+    // mask + scale; per-q tile max (this lane holds keys k0+16t+g*4+r)
+    float sv[4][4];
+    float tmax = -1e30f;
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int key = k0 + t * 16 + g * 4 + r;
+        const bool valid = key < slen && key < S;
+        sv[t][r] = valid ? s_acc[t][r] * scale : -1e30f;
+        tmax = fmaxf(tmax, sv[t][r]);
+      }
+    }
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    const float m_new = fmaxf(m_run, tmax);
+    const float alpha = __expf(m_run - m_new);
+    float rowsum = 0.f;
+#pragma unroll
+    for (int t = 0; t < 4; ++t)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        sv[t][r] = __expf(sv[t][r] - m_new);
+        rowsum += sv[t][r];
+      }
+    rowsum += __shfl_xor(rowsum, 16, 64);
+    rowsum += __shfl_xor(rowsum, 32, 64);
+    l_run = l_run * alpha + rowsum;
+    m_run = m_new;
+
+    // O rescale: alpha indexed by this wave's q rows via LDS broadcast
+    if (g == 0) alpha_lds[wave * 16 + li] = alpha;
+    __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0): same-wave LDS visibility
+#pragma unroll
+    for (int n = 0; n < 4; ++n)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        acc_o[n][r] *= alpha_lds[wave * 16 + g * 4 + r];
+
+    // dropout on the unnormalized probabilities (l keeps the full sum)
+    if (TRAIN_DROP) {
+      const int q_abs = q_row;
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+#pragma unroll
+        for (int rpair = 0; rpair < 2; ++rpair) {
+          const int key = k0 + t * 16 + g * 4 + rpair * 2;
+          uint32_t r4[4];
+          philox(drop_base + static_cast<uint64_t>(q_abs >> 1) * s2 + (key >> 1),
+                 r4);
+          const int w0 = (q_abs & 1) * 2;
+#pragma unroll
+          for (int j = 0; j < 2; ++j) {
+            const bool keep = u32_to_uniform(r4[w0 + j]) >= p;
+            sv[t][rpair * 2 + j] =
+                keep ? sv[t][rpair * 2 + j] * inv_keep : 0.f;
+          }
+        }
+      }
+    }
+
+    // P fragments chain from sv: chunk c covers keys 32c..32c+31
+    bf16x8 pa[2];
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      union {
+        bf16x8 v;
+        __bf16 e[8];
+      } pk;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        pk.e[e] = __bf16(sv[2 * c + (e >> 2)][e & 3]);
+      pa[c] = pk.v;
+    }
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      const __bf16* vrow = &Vt_lds[(n * 16 + li) * kStride];
+      acc_o[n] = MFMA16(pa[0], frag_row(vrow, 0, g), acc_o[n]);
+      acc_o[n] = MFMA16(pa[1], frag_row(vrow, 32, g), acc_o[n]);
+    }
+  }
+
+  // epilogue: normalize rows by l (broadcast per-wave through LDS)
+  if (g == 0) stat_lds[wave * 16 + li] = l_run;
+  __builtin_amdgcn_s_waitcnt(0);
+#pragma unroll
+  for (int n = 0; n < 4; ++n) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qr = q0 + wave * 16 + g * 4 + r;
+      if (qr < S) {
+        const float l = stat_lds[wave * 16 + g * 4 + r];
+        const float o = acc_o[n][r] / (l > 0.f ? l : 1.f);
+        out[(static_cast<int64_t>(b) * S + qr) * H + h * 64 + n * 16 + li] =
+            __bf16(o);
+      }
+    }
+  }
+  if (g == 0 && wave * 16 + li + q0 < S && q_row < S)
+    lse_out[static_cast<int64_t>(bh) * S + q_row] =
+        m_run + __logf(l_run > 0.f ? l_run : 1.f);
+}
+
+// ---------------------------------------------------------------------------
+// backward
+// ---------------------------------------------------------------------------
+// delta[bh][q] = rowsum(dO * O); one wave per row (DH = 64 lanes)
+__global__ void attn_delta_kernel(const __bf16* __restrict__ dout,
+                                  const __bf16* __restrict__ out,
+                                  float* __restrict__ delta, int B, int S,
+                                  int NH) {
+  const int H = NH * 64;
+  const int64_t rows = static_cast<int64_t>(B) * NH * S;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t row = static_cast<int64_t>(blockIdx.x) * 4 + wave;
+  if (row >= rows) return;
+  // row index: bh = row / S, q = row % S
+  const int64_t bh = row / S;
+  const int q = static_cast<int>(row % S);
+  const int b = static_cast<int>(bh) / NH, h = static_cast<int>(bh) % NH;
+  const int64_t base = (static_cast<int64_t>(b) * S + q) * H + h * 64 + lane;
+  const float d = __bfloat162float(dout[base]) * __bfloat162float(out[base]);
+  delta[row] = wave_reduce_sum(d);
+}
+
+template <bool TRAIN_DROP>
+__global__ __launch_bounds__(256) void attn_bwd_kernel(
+    const __bf16* __restrict__ dout, const __bf16* __restrict__ qkv,
+    const int* __restrict__ seqlens, const float* __restrict__ lse,
+    const float* __restrict__ delta, __bf16* __restrict__ dqkv,
+    float* __restrict__ dq_ws, int B, int S, int NH, float p, float scale,
+    uint64_t seed, uint64_t offset) {
+  const int bh = blockIdx.y;
+  const int b = bh / NH, h = bh % NH;
+  const int k0 = blockIdx.x * 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int g = (lane >> 4), li = lane & 15;
+  const int H = NH * 64;
+  const int rs3 = 3 * H;
+  const __bf16* qbase = qkv + static_cast<int64_t>(b) * S * rs3 + h * 64;
+  const __bf16* kbase = qbase + H;
+  const __bf16* vbase = qbase + 2 * H;
+  const __bf16* dobase = dout + static_cast<int64_t>(b) * S * H + h * 64;
+  const int slen = seqlens[b];
+  const float inv_keep = TRAIN_DROP ? 1.f / (1.f - p) : 1.f;
+  Philox philox(seed);
+  const int s2 = S >> 1;
+  const uint64_t drop_base = offset + static_cast<uint64_t>(bh) * s2 * s2;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* K_lds = reinterpret_cast<__bf16*>(smem);   // [64][72] natural
+  __bf16* Kt_lds = K_lds + 64 * kStride;             // [64][72] transposed
+  __bf16* V_lds = Kt_lds + 64 * kStride;             // [64][72] natural
+  __bf16* Q_lds = V_lds + 64 * kStride;              // [64][72] natural
+  __bf16* Qt_lds = Q_lds + 64 * kStride;             // [64][72] transposed
+  __bf16* dO_lds = Qt_lds + 64 * kStride;            // [64][72] natural
+  __bf16* dOt_lds = dO_lds + 64 * kStride;           // [64][72] transposed
+  __bf16* dS_lds = dOt_lds + 64 * kStride;           // [64][72]
+  float* lse_lds = reinterpret_cast<float*>(dS_lds + 64 * kStride);  // [64]
+  float* dlt_lds = lse_lds + 64;                                     // [64]
+
+  // stage K (natural + transposed) and V (natural) once
+  {
+    const int row = tid >> 2, colc = (tid & 3) * 16;
+    const int krow = k0 + row;
+    __bf16 kv[16], vv[16];
+    if (krow < S) {
+      const uint4* ks =
+          reinterpret_cast<const uint4*>(kbase + static_cast<int64_t>(krow) * rs3 + colc);
+      *reinterpret_cast<uint4*>(&kv[0]) = ks[0];
+      *reinterpret_cast<uint4*>(&kv[8]) = ks[1];
+      const uint4* vs =
+          reinterpret_cast<const uint4*>(vbase + static_cast<int64_t>(krow) * rs3 + colc);
+      *reinterpret_cast<uint4*>(&vv[0]) = vs[0];
+      *reinterpret_cast<uint4*>(&vv[8]) = vs[1];
+    } else {
+#pragma unroll
+      for (int j = 0; j < 16; ++j) kv[j] = vv[j] = __bf16(0.f);
+    }
+    *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc]) =
+        *reinterpret_cast<uint4*>(&kv[0]);
+    *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc + 8]) =
+        *reinterpret_cast<uint4*>(&kv[8]);
+    *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc]) =
+        *reinterpret_cast<uint4*>(&vv[0]);
+    *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc + 8]) =
+        *reinterpret_cast<uint4*>(&vv[8]);
+#pragma unroll
+    for (int j = 0; j < 16; ++j) Kt_lds[(colc + j) * kStride + row] = kv[j];
+  }
+  __syncthreads();
+
+  f32x4 dv_acc[4] = {};  // dV^T[dh-tile][wave's 16 keys]
+  f32x4 dk_acc[4] = {};  // dK^T[d-tile][wave's 16 keys]
+
+  const int n_q = (S + 63) / 64;
+  for (int qt = 0; qt < n_q; ++qt) {
+    const int q0 = qt * 64;
+    __syncthreads();
+    // stage Q, Q^T, dO, dO^T + lse/delta for this q-tile
+    {
+      const int row = tid >> 2, colc = (tid & 3) * 16;
+      const int qrow = q0 + row;
+      __bf16 qv[16], dov[16];
+      if (qrow < S) {
+        const uint4* qs =
+            reinterpret_cast<const uint4*>(qbase + static_cast<int64_t>(qrow) * rs3 + colc);
+        *reinterpret_cast<uint4*>(&qv[0]) = qs[0];
+        *reinterpret_cast<uint4*>(&qv[8]) = qs[1];
+        const uint4* ds =
+            reinterpret_cast<const uint4*>(dobase + static_cast<int64_t>(qrow) * H + colc);
+        *reinterpret_cast<uint4*>(&dov[0]) = ds[0];
+        *reinterpret_cast<uint4*>(&dov[8]) = ds[1];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) qv[j] = dov[j] = __bf16(0.f);
+      }
+      *reinterpret_cast<uint4*>(&Q_lds[row * kStride + colc]) =
+          *reinterpret_cast<uint4*>(&qv[0]);
+      *reinterpret_cast<uint4*>(&Q_lds[row * kStride + colc + 8]) =
+          *reinterpret_cast<uint4*>(&qv[8]);
+      *reinterpret_cast<uint4*>(&dO_lds[row * kStride + colc]) =
+          *reinterpret_cast<uint4*>(&dov[0]);
+      *reinterpret_cast<uint4*>(&dO_lds[row * kStride + colc + 8]) =
+          *reinterpret_cast<uint4*>(&dov[8]);
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        Qt_lds[(colc + j) * kStride + row] = qv[j];
+        dOt_lds[(colc + j) * kStride + row] = dov[j];
+      }
+      if (tid < 64) {
+        const int qr = q0 + tid;
+        lse_lds[tid] = (qr < S) ? lse[static_cast<int64_t>(bh) * S + qr] : 0.f;
+        dlt_lds[tid] = (qr < S) ? delta[static_cast<int64_t>(bh) * S + qr] : 0.f;
+      }
+    }
+    __syncthreads();
+
+    // recompute S tiles for this wave's 16 keys x 64 q-rows:
+    // S[q][key]: A = Q rows, B = K^T (natural K rows); C col = key = li
+    const int key_local = wave * 16 + li;
+    const int key_abs = k0 + key_local;
+    const __bf16* krow_n = &K_lds[key_local * kStride];
+    float pv[4][4];   // P[qtile mq][reg r] for this lane's key
+    float dsv[4][4];  // dS
+#pragma unroll
+    for (int mq = 0; mq < 4; ++mq) {
+      const __bf16* qrow_n = &Q_lds[(mq * 16 + li) * kStride];
+      f32x4 acc = {};
+      acc = MFMA16(frag_row(qrow_n, 0, g), frag_row(krow_n, 0, g), acc);
+      acc = MFMA16(frag_row(qrow_n, 32, g), frag_row(krow_n, 32, g), acc);
+      // dP tile: A = dO rows, B = V^T == natural V rows
+      const __bf16* dorow = &dO_lds[(mq * 16 + li) * kStride];
+      const __bf16* vrow = &V_lds[key_local * kStride];
+      f32x4 dp = {};
+      dp = MFMA16(frag_row(dorow, 0, g), frag_row(vrow, 0, g), dp);
+      dp = MFMA16(frag_row(dorow, 32, g), frag_row(vrow, 32, g), dp);
+      const bool kvalid = key_abs < slen && key_abs < S;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int q_abs = q0 + mq * 16 + g * 4 + r;
+        const float row_lse = lse_lds[mq * 16 + g * 4 + r];
+        const float d_row = dlt_lds[mq * 16 + g * 4 + r];
+        float pr = (kvalid && q_abs < S)
+                       ? __expf(acc[r] * scale - row_lse)
+                       : 0.f;
+        float dpd = dp[r];
+        if (TRAIN_DROP) {
+          uint32_t r4[4];
+          philox(drop_base + static_cast<uint64_t>(q_abs >> 1) * s2 +
+                     (key_abs >> 1),
+                 r4);
+          const bool keep =
+              u32_to_uniform(r4[(q_abs & 1) * 2 + (key_abs & 1)]) >= p;
+          dpd = keep ? dpd * inv_keep : 0.f;
+          pv[mq][r] = keep ? pr * inv_keep : 0.f;  // dropped P feeds dV
+        } else {
+          pv[mq][r] = pr;
+        }
+        dsv[mq][r] = pr * (dpd - d_row) * scale;
+      }
+    }
+
+    // dS -> LDS (for the dQ pass), bf16
+#pragma unroll
+    for (int mq = 0; mq < 4; ++mq)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        dS_lds[(mq * 16 + g * 4 + r) * kStride + key_local] =
+            __bf16(dsv[mq][r]);
+
+    // chain P and dS registers into B-fragments over q (chunk c: q 32c+..)
+    bf16x8 pfrag[2], dsfrag[2];
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      union {
+        bf16x8 v;
+        __bf16 e[8];
+      } a, d2;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        a.e[e] = __bf16(pv[2 * c + (e >> 2)][e & 3]);
+        d2.e[e] = __bf16(dsv[2 * c + (e >> 2)][e & 3]);
+      }
+      pfrag[c] = a.v;
+      dsfrag[c] = d2.v;
+    }
+
+    // dV^T += dO^T x P ; dK^T += Q^T x dS  (per dh/d tile m)
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      const __bf16* dot_row = &dOt_lds[(m * 16 + li) * kStride];
+      const __bf16* qt_row = &Qt_lds[(m * 16 + li) * kStride];
+      dv_acc[m] = MFMA16(frag_row(dot_row, 0, g), pfrag[0], dv_acc[m]);
+      dv_acc[m] = MFMA16(frag_row(dot_row, 32, g), pfrag[1], dv_acc[m]);
+      dk_acc[m] = MFMA16(frag_row(qt_row, 0, g), dsfrag[0], dk_acc[m]);
+      dk_acc[m] = MFMA16(frag_row(qt_row, 32, g), dsfrag[1], dk_acc[m]);
+    }
+    __syncthreads();  // dS_lds complete across waves
+
+    // dQ: wave owns q-subtile `wave`; A = dS rows, B = K^T (Kt rows)
+    {
+      const __bf16* ds_row = &dS_lds[(wave * 16 + li) * kStride];
+      bf16x8 a0 = frag_row(ds_row, 0, g);
+      bf16x8 a1 = frag_row(ds_row, 32, g);
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        const __bf16* kt_row = &Kt_lds[(n * 16 + li) * kStride];
+        f32x4 dq = {};
+        dq = MFMA16(a0, frag_row(kt_row, 0, g), dq);
+        dq = MFMA16(a1, frag_row(kt_row, 32, g), dq);
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int q_abs = q0 + wave * 16 + g * 4 + r;
+          if (q_abs < S)
+            atomicAdd(&dq_ws[(static_cast<int64_t>(bh) * S + q_abs) * 64 +
+                             n * 16 + li],
+                      dq[r]);
+        }
+      }
+    }
+  }
+
+  // write dK/dV straight into dqkv (this block owns keys k0..k0+63)
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int dh = m * 16 + g * 4 + r;  // C row = dh/d index
+      const int key_abs = k0 + wave * 16 + li;
+      if (key_abs < S) {
+        const int64_t rowb = static_cast<int64_t>(b) * S + key_abs;
+        dqkv[rowb * rs3 + H + h * 64 + dh] = __bf16(dk_acc[m][r]);
+        dqkv[rowb * rs3 + 2 * H + h * 64 + dh] = __bf16(dv_acc[m][r]);
+      }
+    }
+  }
+}
+
+// pack fp32 dQ workspace into the bf16 dqkv Q slots
+__global__ void attn_pack_dq_kernel(const float* __restrict__ dq_ws,
+                                    __bf16* __restrict__ dqkv, int B, int S,
+                                    int NH) {
+  const int H = NH * 64;
+  const int64_t total = static_cast<int64_t>(B) * NH * S * 64;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += static_cast<int64_t>(gridDim.x) * blockDim.x) {
+    const int d = static_cast<int>(i & 63);
+    const int64_t rest = i >> 6;
+    const int q = static_cast<int>(rest % S);
+    const int64_t bh = rest / S;
+    const int b = static_cast<int>(bh) / NH, h = static_cast<int>(bh) % NH;
+    dqkv[(static_cast<int64_t>(b) * S + q) * 3 * H + h * 64 + d] =
+        __bf16(dq_ws[i]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> attention_fwd(torch::Tensor qkv,
+                                         torch::Tensor seqlens,
+                                         int64_t num_heads, double p,
+                                         int64_t seed, int64_t offset) {
+  TORCH_CHECK(qkv.dim() == 3 && qkv.is_contiguous(), "attn_fwd: bad qkv");
+  TORCH_CHECK(qkv.scalar_type() == torch::kBFloat16,
+              "attn_fwd: bf16 only (autocast path)");
+  const int B = qkv.size(0), S = qkv.size(1);
+  const int H = qkv.size(2) / 3;
+  const int NH = static_cast<int>(num_heads);
+  TORCH_CHECK(H == NH * 64, "attn_fwd: head_dim must be 64");
+  TORCH_CHECK(S % 16 == 0, "attn_fwd: S must be a multiple of 16");
+  auto seql = seqlens.to(qkv.device(), torch::kInt32).contiguous();
+  auto out = torch::empty({B, S, H}, qkv.options());
+  auto lse = torch::empty({B * NH, S}, qkv.options().dtype(torch::kFloat32));
+  const float scale = 1.0f / sqrtf(64.f);
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 grid((S + 63) / 64, B * NH), block(256);
+  const size_t lds = 2 * 64 * kStride * sizeof(__bf16) + 2 * 64 * sizeof(float);
+  const bool train_drop = p > 0.0;
+  if (train_drop) {
+    hipLaunchKernelGGL((attn_fwd_kernel<true>), grid, block, lds, stream,
+                       reinterpret_cast<const __bf16*>(qkv.data_ptr()),
+                       seql.data_ptr<int>(),
+                       reinterpret_cast<__bf16*>(out.data_ptr()),
+                       lse.data_ptr<float>(), B, S, NH, static_cast<float>(p),
+                       scale, static_cast<uint64_t>(seed),
+                       static_cast<uint64_t>(offset));
+  } else {
+    hipLaunchKernelGGL((attn_fwd_kernel<false>), grid, block, lds, stream,
+                       reinterpret_cast<const __bf16*>(qkv.data_ptr()),
+                       seql.data_ptr<int>(),
+                       reinterpret_cast<__bf16*>(out.data_ptr()),
+                       lse.data_ptr<float>(), B, S, NH, static_cast<float>(p),
+                       scale, static_cast<uint64_t>(seed),
+                       static_cast<uint64_t>(offset));
+  }
+  return {out, lse};
+}
+
+torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
+                            torch::Tensor seqlens, torch::Tensor out,
+                            torch::Tensor lse, int64_t num_heads, double p,
+                            int64_t seed, int64_t offset) {
+  const int B = qkv.size(0), S = qkv.size(1);
+  const int H = qkv.size(2) / 3;
+  const int NH = static_cast<int>(num_heads);
+  auto seql = seqlens.to(qkv.device(), torch::kInt32).contiguous();
+  auto dqkv = torch::empty_like(qkv);
+  auto fopts = qkv.options().dtype(torch::kFloat32);
+  auto dq_ws = torch::zeros({static_cast<int64_t>(B) * NH * S * 64}, fopts);
+  auto delta = torch::empty({static_cast<int64_t>(B) * NH * S}, fopts);
+  auto stream = at::hip::getCurrentHIPStream();
+  auto dout_c = dout.contiguous();
+
+  const int64_t rows = static_cast<int64_t>(B) * NH * S;
+  hipLaunchKernelGGL(attn_delta_kernel, dim3((rows + 3) / 4), dim3(256), 0,
+                     stream,
+                     reinterpret_cast<const __bf16*>(dout_c.data_ptr()),
+                     reinterpret_cast<const __bf16*>(out.data_ptr()),
+                     delta.data_ptr<float>(), B, S, NH);
+
+  dim3 grid((S + 63) / 64, B * NH), block(256);
+  const size_t lds = 8 * 64 * kStride * sizeof(__bf16) + 2 * 64 * sizeof(float);
+  // backward wants ~74 KB LDS (> the 64 KB default cap; MI355X has 160)
+  HIP_CHECK(hipFuncSetAttribute(
+      reinterpret_cast<const void*>(&attn_bwd_kernel<true>),
+      hipFuncAttributeMaxDynamicSharedMemorySize, lds));
+  HIP_CHECK(hipFuncSetAttribute(
+      reinterpret_cast<const void*>(&attn_bwd_kernel<false>),
+      hipFuncAttributeMaxDynamicSharedMemorySize, lds));
+  const float scale = 1.0f / sqrtf(64.f);
+  const bool train_drop = p > 0.0;
+  if (train_drop) {
+    hipLaunchKernelGGL((attn_bwd_kernel<true>), grid, block, lds, stream,
+                       reinterpret_cast<const __bf16*>(dout_c.data_ptr()),
+                       reinterpret_cast<const __bf16*>(qkv.data_ptr()),
+                       seql.data_ptr<int>(), lse.data_ptr<float>(),
+                       delta.data_ptr<float>(),
+                       reinterpret_cast<__bf16*>(dqkv.data_ptr()),
+                       dq_ws.data_ptr<float>(), B, S, NH,
+                       static_cast<float>(p), scale,
+                       static_cast<uint64_t>(seed),
+                       static_cast<uint64_t>(offset));
+  } else {
+    hipLaunchKernelGGL((attn_bwd_kernel<false>), grid, block, lds, stream,
+                       reinterpret_cast<const __bf16*>(dout_c.data_ptr()),
+                       reinterpret_cast<const __bf16*>(qkv.data_ptr()),
+                       seql.data_ptr<int>(), lse.data_ptr<float>(),
+                       delta.data_ptr<float>(),
+                       reinterpret_cast<__bf16*>(dqkv.data_ptr()),
+                       dq_ws.data_ptr<float>(), B, S, NH,
+                       static_cast<float>(p), scale,
+                       static_cast<uint64_t>(seed),
+                       static_cast<uint64_t>(offset));
+  }
+  const int64_t total = rows * 64;
+  const int pblocks = static_cast<int>(std::min<int64_t>((total + 255) / 256, 2048));
+  hipLaunchKernelGGL(attn_pack_dq_kernel, dim3(pblocks), dim3(256), 0, stream,
+                     dq_ws.data_ptr<float>(),
+                     reinterpret_cast<__bf16*>(dqkv.data_ptr()), B, S, NH);
+  return dqkv;
+}
+
+}  // namespace bpa

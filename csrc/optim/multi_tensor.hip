@@ -1,0 +1,298 @@
+// Multi-tensor optimizer kernels for MI355X (gfx950).
+//
+// Replaces the Apex amp_C / FusedLAMB / FusedAdam CUDA kernels the
+// reference imports (src/optimization.py:25-33, run_pretraining.py:39):
+//   * multi_tensor_l2norm_sq : global grad-norm^2 over a tensor list
+//   * multi_tensor_clip_scale: in-place clip by global norm
+//   * fused_lamb             : stage1 (moments+update+per-tensor norms,
+//     update overwrites the grad buffer) + stage2 (trust-ratio apply)
+//   * fused_adam             : one-stage AdamW/Adam apply
+//
+// A chunk table ([tensor_idx, element_offset] per 64Ki-element chunk)
+// and per-tensor pointer tables are built host-side each call and
+// copied once; all kernels then run without host synchronization (the
+// LAMB trust ratio and the clip factor are read on-device).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "../common.h"
+
+namespace bpa {
+
+constexpr int kChunk = 65536;
+constexpr int kThreads = 256;
+
+namespace {
+
+struct ChunkedList {
+  torch::Tensor ptrs;    // [n_lists, n_tensors] int64 (device)
+  torch::Tensor chunks;  // [n_chunks, 3] int32: tensor, offset_lo, offset_hi
+  int n_chunks;
+};
+
+ChunkedList build_chunks(const std::vector<std::vector<torch::Tensor>>& lists,
+                         const torch::Device& device) {
+  const int n_lists = lists.size();
+  const int n = lists[0].size();
+  auto cpu_i64 = torch::TensorOptions().dtype(torch::kLong);
+  auto cpu_i32 = torch::TensorOptions().dtype(torch::kInt);
+  auto ptrs_cpu = torch::empty({n_lists, n}, cpu_i64);
+  auto* pp = ptrs_cpu.data_ptr<int64_t>();
+  std::vector<int> ct, clo, chi;
+  for (int t = 0; t < n; ++t) {
+    const int64_t numel = lists[0][t].numel();
+    for (int l = 0; l < n_lists; ++l) {
+      TORCH_CHECK(lists[l][t].is_contiguous() &&
+                      lists[l][t].scalar_type() == torch::kFloat32,
+                  "multi_tensor: fp32 contiguous tensors required");
+      TORCH_CHECK(lists[l][t].numel() == numel, "multi_tensor: size mismatch");
+      pp[l * n + t] = reinterpret_cast<int64_t>(lists[l][t].data_ptr());
+    }
+    for (int64_t off = 0; off < numel; off += kChunk) {
+      ct.push_back(t);
+      clo.push_back(static_cast<int>(off & 0xFFFFFFFF));
+      chi.push_back(static_cast<int>(off >> 32));
+    }
+  }
+  const int n_chunks = ct.size();
+  auto chunks_cpu = torch::empty({n_chunks, 3}, cpu_i32);
+  auto* cp = chunks_cpu.data_ptr<int>();
+  for (int i = 0; i < n_chunks; ++i) {
+    cp[i * 3] = ct[i];
+    cp[i * 3 + 1] = clo[i];
+    cp[i * 3 + 2] = chi[i];
+  }
+  return {ptrs_cpu.to(device), chunks_cpu.to(device), n_chunks};
+}
+
+__device__ __forceinline__ int64_t chunk_off(const int* info) {
+  return (static_cast<int64_t>(info[2]) << 32) |
+         static_cast<uint32_t>(info[1]);
+}
+
+}  // namespace
+
+// sizes come along as a device tensor [n] int64
+__global__ void l2norm_sq_kernel(const int64_t* __restrict__ ptrs,
+                                 const int64_t* __restrict__ sizes,
+                                 const int* __restrict__ chunks,
+                                 float* __restrict__ out, int n_chunks) {
+  const int chunk = blockIdx.x;
+  if (chunk >= n_chunks) return;
+  const int t = chunks[chunk * 3];
+  const int64_t off = chunk_off(&chunks[chunk * 3]);
+  const float* g = reinterpret_cast<const float*>(ptrs[t]) + off;
+  const int n = static_cast<int>(tmin<int64_t>(kChunk, sizes[t] - off));
+  float acc = 0.f;
+  for (int i = threadIdx.x; i < n; i += blockDim.x) acc += g[i] * g[i];
+  __shared__ float smem[kThreads / WAVE_SIZE];
+  acc = block_reduce_sum<kThreads / WAVE_SIZE>(acc, smem);
+  if (threadIdx.x == 0) atomicAdd(out, acc);
+}
+
+__global__ void clip_scale_kernel(const int64_t* __restrict__ ptrs,
+                                  const int64_t* __restrict__ sizes,
+                                  const int* __restrict__ chunks,
+                                  const float* __restrict__ gnorm_sq,
+                                  float max_norm, int n_chunks) {
+  const int chunk = blockIdx.x;
+  if (chunk >= n_chunks) return;
+  const float gn = sqrtf(*gnorm_sq);
+  const float scale = (max_norm > 0.f && gn > max_norm) ? max_norm / gn : 1.f;
+  if (scale == 1.f) return;
+  const int t = chunks[chunk * 3];
+  const int64_t off = chunk_off(&chunks[chunk * 3]);
+  float* g = reinterpret_cast<float*>(ptrs[t]) + off;
+  const int n = static_cast<int>(tmin<int64_t>(kChunk, sizes[t] - off));
+  for (int i = threadIdx.x; i < n; i += blockDim.x) g[i] *= scale;
+}
+
+// LAMB stage 1: m/v update, bias-corrected Adam direction (+wd*p) written
+// over the grad buffer; per-tensor ||p||^2 and ||u||^2 into norms[t][2].
+__global__ void lamb_stage1_kernel(
+    const int64_t* __restrict__ ptrs,  // [4][n]: p, g, m, v
+    const int64_t* __restrict__ sizes, const int* __restrict__ chunks,
+    const float* __restrict__ gnorm_sq, float* __restrict__ norms,
+    int n_tensors, int n_chunks, float beta1, float beta2, float eps,
+    float wd, float bc1, float bc2, float beta1_g, float max_grad_norm) {
+  const int chunk = blockIdx.x;
+  if (chunk >= n_chunks) return;
+  const int t = chunks[chunk * 3];
+  const int64_t off = chunk_off(&chunks[chunk * 3]);
+  float* p = reinterpret_cast<float*>(ptrs[t]) + off;
+  float* g = reinterpret_cast<float*>(ptrs[n_tensors + t]) + off;
+  float* m = reinterpret_cast<float*>(ptrs[2 * n_tensors + t]) + off;
+  float* v = reinterpret_cast<float*>(ptrs[3 * n_tensors + t]) + off;
+  const int n = static_cast<int>(tmin<int64_t>(kChunk, sizes[t] - off));
+
+  const float gn = sqrtf(*gnorm_sq);
+  const float clip =
+      (max_grad_norm > 0.f && gn > max_grad_norm) ? max_grad_norm / gn : 1.f;
+
+  float pn = 0.f, un = 0.f;
+  for (int i = threadIdx.x; i < n; i += blockDim.x) {
+    const float gi = g[i] * clip;
+    const float mi = beta1 * m[i] + beta1_g * gi;
+    const float vi = beta2 * v[i] + (1.f - beta2) * gi * gi;
+    m[i] = mi;
+    v[i] = vi;
+    float u = (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    u += wd * p[i];
+    g[i] = u;  // update overwrites grad
+    pn += p[i] * p[i];
+    un += u * u;
+  }
+  __shared__ float smem[kThreads / WAVE_SIZE];
+  pn = block_reduce_sum<kThreads / WAVE_SIZE>(pn, smem);
+  __syncthreads();
+  un = block_reduce_sum<kThreads / WAVE_SIZE>(un, smem);
+  if (threadIdx.x == 0) {
+    atomicAdd(&norms[t * 2], pn);
+    atomicAdd(&norms[t * 2 + 1], un);
+  }
+}
+
+__global__ void lamb_stage2_kernel(const int64_t* __restrict__ ptrs,
+                                   const int64_t* __restrict__ sizes,
+                                   const int* __restrict__ chunks,
+                                   const float* __restrict__ norms,
+                                   int n_tensors, int n_chunks, float lr,
+                                   bool use_ratio) {
+  const int chunk = blockIdx.x;
+  if (chunk >= n_chunks) return;
+  const int t = chunks[chunk * 3];
+  const int64_t off = chunk_off(&chunks[chunk * 3]);
+  float* p = reinterpret_cast<float*>(ptrs[t]) + off;
+  const float* u = reinterpret_cast<const float*>(ptrs[n_tensors + t]) + off;
+  const int n = static_cast<int>(tmin<int64_t>(kChunk, sizes[t] - off));
+  float ratio = 1.f;
+  if (use_ratio) {
+    const float wn = sqrtf(norms[t * 2]);
+    const float un = sqrtf(norms[t * 2 + 1]);
+    if (wn > 0.f && un > 0.f) ratio = wn / un;
+  }
+  const float step_size = lr * ratio;
+  for (int i = threadIdx.x; i < n; i += blockDim.x) p[i] -= step_size * u[i];
+}
+
+__global__ void adam_kernel(const int64_t* __restrict__ ptrs,
+                            const int64_t* __restrict__ sizes,
+                            const int* __restrict__ chunks, int n_tensors,
+                            int n_chunks, float lr, float beta1, float beta2,
+                            float eps, float wd, float bc1, float bc2,
+                            bool adam_w) {
+  const int chunk = blockIdx.x;
+  if (chunk >= n_chunks) return;
+  const int t = chunks[chunk * 3];
+  const int64_t off = chunk_off(&chunks[chunk * 3]);
+  float* p = reinterpret_cast<float*>(ptrs[t]) + off;
+  const float* g = reinterpret_cast<const float*>(ptrs[n_tensors + t]) + off;
+  float* m = reinterpret_cast<float*>(ptrs[2 * n_tensors + t]) + off;
+  float* v = reinterpret_cast<float*>(ptrs[3 * n_tensors + t]) + off;
+  const int n = static_cast<int>(tmin<int64_t>(kChunk, sizes[t] - off));
+  for (int i = threadIdx.x; i < n; i += blockDim.x) {
+    float gi = g[i];
+    if (!adam_w && wd != 0.f) gi += wd * p[i];  // L2 mode
+    const float mi = beta1 * m[i] + (1.f - beta1) * gi;
+    const float vi = beta2 * v[i] + (1.f - beta2) * gi * gi;
+    m[i] = mi;
+    v[i] = vi;
+    float u = (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    if (adam_w && wd != 0.f) u += wd * p[i];
+    p[i] -= lr * u;
+  }
+}
+
+static torch::Tensor sizes_tensor(const std::vector<torch::Tensor>& ts,
+                                  const torch::Device& device) {
+  auto cpu = torch::empty({static_cast<int64_t>(ts.size())},
+                          torch::TensorOptions().dtype(torch::kLong));
+  auto* sp = cpu.data_ptr<int64_t>();
+  for (size_t i = 0; i < ts.size(); ++i) sp[i] = ts[i].numel();
+  return cpu.to(device);
+}
+
+torch::Tensor multi_tensor_l2norm_sq(std::vector<torch::Tensor> tensors) {
+  TORCH_CHECK(!tensors.empty(), "l2norm: empty list");
+  auto device = tensors[0].device();
+  auto meta = build_chunks({tensors}, device);
+  auto sizes = sizes_tensor(tensors, device);
+  auto out = torch::zeros({1}, tensors[0].options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(l2norm_sq_kernel, dim3(meta.n_chunks), dim3(kThreads), 0,
+                     stream, meta.ptrs.data_ptr<int64_t>(),
+                     sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(),
+                     out.data_ptr<float>(), meta.n_chunks);
+  return out.squeeze(0);
+}
+
+void multi_tensor_clip_scale(std::vector<torch::Tensor> grads,
+                             torch::Tensor gnorm_sq, double max_norm) {
+  auto device = grads[0].device();
+  auto meta = build_chunks({grads}, device);
+  auto sizes = sizes_tensor(grads, device);
+  auto gsq = gnorm_sq.reshape({1}).contiguous();
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(clip_scale_kernel, dim3(meta.n_chunks), dim3(kThreads), 0,
+                     stream, meta.ptrs.data_ptr<int64_t>(),
+                     sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(),
+                     gsq.data_ptr<float>(), static_cast<float>(max_norm),
+                     meta.n_chunks);
+}
+
+void fused_lamb(std::vector<torch::Tensor> params,
+                std::vector<torch::Tensor> grads,
+                std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
+                torch::Tensor gnorm_sq, double lr, double beta1, double beta2,
+                double eps, double wd, int64_t step, bool bias_correction,
+                bool grad_averaging, double max_grad_norm, bool use_ratio) {
+  const int n = params.size();
+  TORCH_CHECK(n > 0, "fused_lamb: empty");
+  auto device = params[0].device();
+  auto meta = build_chunks({params, grads, ms, vs}, device);
+  auto sizes = sizes_tensor(params, device);
+  auto norms = torch::zeros({n, 2}, params[0].options().dtype(torch::kFloat32));
+  auto gsq = gnorm_sq.reshape({1}).contiguous();
+  const float bc1 = bias_correction ? 1.f - powf(beta1, step) : 1.f;
+  const float bc2 = bias_correction ? 1.f - powf(beta2, step) : 1.f;
+  const float beta1_g = grad_averaging ? 1.f - beta1 : 1.f;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(lamb_stage1_kernel, dim3(meta.n_chunks), dim3(kThreads),
+                     0, stream, meta.ptrs.data_ptr<int64_t>(),
+                     sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(),
+                     gsq.data_ptr<float>(), norms.data_ptr<float>(), n,
+                     meta.n_chunks, static_cast<float>(beta1),
+                     static_cast<float>(beta2), static_cast<float>(eps),
+                     static_cast<float>(wd), bc1, bc2, beta1_g,
+                     static_cast<float>(max_grad_norm));
+  hipLaunchKernelGGL(lamb_stage2_kernel, dim3(meta.n_chunks), dim3(kThreads),
+                     0, stream, meta.ptrs.data_ptr<int64_t>(),
+                     sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(),
+                     norms.data_ptr<float>(), n, meta.n_chunks,
+                     static_cast<float>(lr), use_ratio);
+}
+
+void fused_adam(std::vector<torch::Tensor> params,
+                std::vector<torch::Tensor> grads,
+                std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
+                double lr, double beta1, double beta2, double eps, double wd,
+                int64_t step, bool bias_correction, bool adam_w) {
+  const int n = params.size();
+  TORCH_CHECK(n > 0, "fused_adam: empty");
+  auto device = params[0].device();
+  auto meta = build_chunks({params, grads, ms, vs}, device);
+  auto sizes = sizes_tensor(params, device);
+  const float bc1 = bias_correction ? 1.f - powf(beta1, step) : 1.f;
+  const float bc2 = bias_correction ? 1.f - powf(beta2, step) : 1.f;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(adam_kernel, dim3(meta.n_chunks), dim3(kThreads), 0,
+                     stream, meta.ptrs.data_ptr<int64_t>(),
+                     sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(), n,
+                     meta.n_chunks, static_cast<float>(lr),
+                     static_cast<float>(beta1), static_cast<float>(beta2),
+                     static_cast<float>(eps), static_cast<float>(wd), bc1, bc2,
+                     adam_w);
+}
+
+}  // namespace bpa

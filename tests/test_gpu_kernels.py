@@ -1,0 +1,422 @@
+"""HIP kernel numerics: every gfx950 kernel vs the eager fp32 reference
+(the same composites the CPU path runs). All tests @pytest.mark.gpu."""
+
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from bert_pytorch_amd import ops
+    from bert_pytorch_amd.ops import _reference as ref
+
+DEV = "cuda:0"
+
+
+def rel_err(a: torch.Tensor, b: torch.Tensor) -> float:
+    a = a.float()
+    b = b.float()
+    denom = b.abs().max().clamp(min=1e-3)
+    return float((a - b).abs().max() / denom)
+
+
+def ext():
+    return ops.extension()
+
+
+# ---------------------------------------------------------------------------
+# LayerNorm
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("shape", [(128, 1024), (777, 1024), (64, 4096)])
+def test_ln_forward(dtype, shape):
+    torch.manual_seed(0)
+    x = torch.randn(shape, device=DEV, dtype=dtype)
+    w = torch.randn(shape[1], device=DEV) * 0.1 + 1.0
+    b = torch.randn(shape[1], device=DEV) * 0.1
+    y, mean, rstd = ext().ln_fwd(x, w, b, 1e-12)
+    y_ref = ref.layer_norm(x.float(), w, b, 1e-12)
+    tol = 1e-5 if dtype == torch.float32 else 2e-2
+    assert rel_err(y, y_ref) < tol
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_ln_backward(dtype):
+    torch.manual_seed(1)
+    rows, H = 512, 1024
+    x = torch.randn(rows, H, device=DEV, dtype=dtype)
+    w = (torch.randn(H, device=DEV) * 0.1 + 1.0).requires_grad_(True)
+    b = (torch.randn(H, device=DEV) * 0.1).requires_grad_(True)
+    xr = x.float().detach().requires_grad_(True)
+    y_ref = ref.layer_norm(xr, w, b, 1e-12)
+    dy = torch.randn_like(y_ref)
+    y_ref.backward(dy)
+
+    _, mean, rstd = ext().ln_fwd(x, w.detach(), b.detach(), 1e-12)
+    dx, dw, db = ext().ln_bwd(dy.to(dtype), x, w.detach(), mean, rstd)
+    tol = 1e-4 if dtype == torch.float32 else 3e-2
+    assert rel_err(dx, xr.grad) < tol
+    assert rel_err(dw, w.grad) < tol
+    assert rel_err(db, b.grad) < tol
+
+
+# ---------------------------------------------------------------------------
+# bias + GELU
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_bias_gelu(dtype):
+    torch.manual_seed(2)
+    rows, H = 512, 4096
+    x = torch.randn(rows, H, device=DEV, dtype=dtype)
+    bias = torch.randn(H, device=DEV, requires_grad=True)
+    xr = x.float().detach().requires_grad_(True)
+    y_ref = ref.bias_gelu(xr, bias)
+    dy = torch.randn_like(y_ref)
+    y_ref.backward(dy)
+
+    y = ext().bias_gelu_fwd(x, bias.detach())
+    dx, db = ext().bias_gelu_bwd(dy.to(dtype), x, bias.detach())
+    tol = 1e-5 if dtype == torch.float32 else 2e-2
+    assert rel_err(y, y_ref) < tol
+    assert rel_err(dx, xr.grad) < (1e-4 if dtype == torch.float32 else 3e-2)
+    assert rel_err(db, bias.grad) < (1e-4 if dtype == torch.float32 else 3e-2)
+
+
+# ---------------------------------------------------------------------------
+# fused bias+dropout+residual+LN
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_bdrl_no_dropout(dtype):
+    torch.manual_seed(3)
+    rows, H = 512, 1024
+    x = torch.randn(rows, H, device=DEV, dtype=dtype)
+    res = torch.randn(rows, H, device=DEV, dtype=dtype)
+    bias = torch.randn(H, device=DEV, requires_grad=True)
+    w = (torch.randn(H, device=DEV) * 0.1 + 1.0).requires_grad_(True)
+    lb = (torch.randn(H, device=DEV) * 0.1).requires_grad_(True)
+    xr = x.float().detach().requires_grad_(True)
+    rr = res.float().detach().requires_grad_(True)
+    y_ref = ref.bias_dropout_residual_ln(xr, bias, rr, w, lb, 0.0, False)
+    dy = torch.randn_like(y_ref)
+    y_ref.backward(dy)
+
+    y, z, mask, mean, rstd = ext().bias_dropout_residual_ln_fwd(
+        x, bias.detach(), res, w.detach(), lb.detach(), 0.0, 1e-12, 0, 0
+    )
+    assert rel_err(y, y_ref) < (1e-5 if dtype == torch.float32 else 2e-2)
+    dx, db, dres, dw, dlb = ext().bias_dropout_residual_ln_bwd(
+        dy.to(dtype), z, mask, w.detach(), mean, rstd, 0.0, True
+    )
+    tol = 1e-4 if dtype == torch.float32 else 3e-2
+    assert rel_err(dx, xr.grad) < tol
+    assert rel_err(dres, rr.grad) < tol
+    assert rel_err(db, bias.grad) < tol
+    assert rel_err(dw, w.grad) < tol
+    assert rel_err(dlb, lb.grad) < tol
+
+
+def test_bdrl_dropout_self_consistent():
+    """With p>0: y must equal LN(mask*(x+b)/(1-p)+res) for the returned
+    mask, drop fraction ~= p, and backward consistent with the mask."""
+    torch.manual_seed(4)
+    rows, H, p = 1024, 1024, 0.3
+    x = torch.randn(rows, H, device=DEV, dtype=torch.bfloat16)
+    res = torch.randn(rows, H, device=DEV, dtype=torch.bfloat16)
+    bias = torch.randn(H, device=DEV)
+    w = torch.randn(H, device=DEV) * 0.1 + 1.0
+    lb = torch.randn(H, device=DEV) * 0.1
+    y, z, mask, mean, rstd = ext().bias_dropout_residual_ln_fwd(
+        x, bias, res, w, lb, p, 1e-12, 12345, 7
+    )
+    frac = float(mask.float().mean())
+    assert abs(frac - (1 - p)) < 0.01
+    # recompute with the mask through the fp32 composite
+    keep = mask.float()
+    z_ref = keep * (x.float() + bias) / (1 - p) + res.float()
+    assert rel_err(z, z_ref) < 2e-2
+    y_ref = ref.layer_norm(z_ref, w, lb, 1e-12)
+    assert rel_err(y, y_ref) < 3e-2
+
+
+# ---------------------------------------------------------------------------
+# embedding + LN + dropout
+# ---------------------------------------------------------------------------
+def test_embedding_fwd_bwd():
+    torch.manual_seed(5)
+    B, S, H, V, P, T = 8, 128, 1024, 1000, 512, 2
+    ids = torch.randint(0, V, (B, S), device=DEV)
+    tt = torch.randint(0, T, (B, S), device=DEV)
+    word = torch.randn(V, H, device=DEV, requires_grad=True)
+    pos = torch.randn(P, H, device=DEV, requires_grad=True)
+    tok = torch.randn(T, H, device=DEV, requires_grad=True)
+    w = (torch.randn(H, device=DEV) * 0.1 + 1.0).requires_grad_(True)
+    lb = (torch.randn(H, device=DEV) * 0.1).requires_grad_(True)
+
+    y_ref = ref.embedding_ln_dropout(ids, tt, word, pos, tok, w, lb, 0.0, False)
+    dy = torch.randn_like(y_ref)
+    y_ref.backward(dy)
+
+    y, z, mask, mean, rstd = ext().embedding_ln_dropout_fwd(
+        ids, tt, word.detach(), pos.detach(), tok.detach(), w.detach(),
+        lb.detach(), 0.0, 1e-12, 0, 0, torch.float32,
+    )
+    assert rel_err(y, y_ref) < 1e-4
+    d_word, d_pos, d_tok, dw, dlb = ext().embedding_ln_dropout_bwd(
+        dy, ids, tt, z, mask, w.detach(), mean, rstd, 0.0, V, P, T
+    )
+    assert rel_err(d_word, word.grad) < 1e-3
+    assert rel_err(d_pos, pos.grad) < 1e-3
+    assert rel_err(d_tok, tok.grad) < 1e-3
+    assert rel_err(dw, w.grad) < 1e-3
+    assert rel_err(dlb, lb.grad) < 1e-3
+
+
+# ---------------------------------------------------------------------------
+# cross entropy
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_cross_entropy(dtype):
+    torch.manual_seed(6)
+    N, V = 512, 30528
+    logits = (torch.randn(N, V, device=DEV) * 3).to(dtype)
+    labels = torch.randint(0, V, (N,), device=DEV)
+    labels[::3] = -1  # ignored rows
+    lr = logits.float().detach().requires_grad_(True)
+    loss_ref = ref.cross_entropy(lr, labels, -1)
+    loss_ref.backward()
+
+    loss = ops.fused_cross_entropy(logits.requires_grad_(True), labels, -1)
+    assert abs(float(loss) - float(loss_ref)) < (
+        1e-4 if dtype == torch.float32 else 3e-2
+    )
+    loss.backward()
+    assert rel_err(logits.grad, lr.grad) < (
+        1e-4 if dtype == torch.float32 else 3e-2
+    )
+
+
+def test_cross_entropy_all_ignored():
+    logits = torch.randn(16, 128, device=DEV)
+    labels = torch.full((16,), -1, device=DEV, dtype=torch.long)
+    loss = ops.fused_cross_entropy(logits, labels, -1)
+    assert float(loss) == 0.0
+
+
+# ---------------------------------------------------------------------------
+# attention
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("B,S,NH", [(4, 128, 16), (2, 512, 16), (3, 128, 4)])
+def test_attention_forward(B, S, NH):
+    torch.manual_seed(7)
+    H = NH * 64
+    qkv = (torch.randn(B, S, 3 * H, device=DEV) * 0.5).bfloat16()
+    seqlens = torch.randint(S // 2, S + 1, (B,), device=DEV, dtype=torch.int32)
+    out, lse = ext().attention_fwd(qkv, seqlens, NH, 0.0, 0, 0)
+    out_ref = ref.attention(qkv.float(), seqlens, NH, 0.0, False)
+    assert rel_err(out, out_ref) < 3e-2
+
+
+def test_attention_forward_spike():
+    """Online-softmax max tracking: one huge key must not break numerics
+    (guide rule 26: force the rescale path)."""
+    torch.manual_seed(8)
+    B, S, NH = 2, 128, 4
+    H = NH * 64
+    qkv = (torch.randn(B, S, 3 * H, device=DEV) * 0.3)
+    # spike a late key so the running max jumps at the last tile
+    qkv[:, S - 3, H : 2 * H] += 40.0
+    qkv = qkv.bfloat16()
+    seqlens = torch.full((B,), S, device=DEV, dtype=torch.int32)
+    out, _ = ext().attention_fwd(qkv, seqlens, NH, 0.0, 0, 0)
+    out_ref = ref.attention(qkv.float(), seqlens, NH, 0.0, False)
+    assert torch.isfinite(out.float()).all()
+    assert rel_err(out, out_ref) < 5e-2
+
+
+@pytest.mark.parametrize("B,S,NH", [(4, 128, 8), (2, 512, 4)])
+def test_attention_backward(B, S, NH):
+    torch.manual_seed(9)
+    H = NH * 64
+    qkv = (torch.randn(B, S, 3 * H, device=DEV) * 0.5).bfloat16()
+    seqlens = torch.randint(S // 2, S + 1, (B,), device=DEV, dtype=torch.int32)
+    qr = qkv.float().detach().requires_grad_(True)
+    out_ref = ref.attention(qr, seqlens, NH, 0.0, False)
+    dout = (torch.randn_like(out_ref) * 0.5)
+    out_ref.backward(dout)
+
+    out, lse = ext().attention_fwd(qkv, seqlens, NH, 0.0, 0, 0)
+    dqkv = ext().attention_bwd(
+        dout.bfloat16(), qkv, seqlens, out, lse, NH, 0.0, 0, 0
+    )
+    assert rel_err(dqkv, qr.grad) < 6e-2
+
+
+def test_attention_dropout_stats_and_determinism():
+    torch.manual_seed(10)
+    B, S, NH, p = 2, 128, 4, 0.5
+    H = NH * 64
+    qkv = (torch.randn(B, S, 3 * H, device=DEV) * 0.5).bfloat16()
+    seqlens = torch.full((B,), S, device=DEV, dtype=torch.int32)
+    o1, l1 = ext().attention_fwd(qkv, seqlens, NH, p, 99, 1)
+    o2, l2 = ext().attention_fwd(qkv, seqlens, NH, p, 99, 1)
+    assert torch.equal(o1, o2), "same philox state must reproduce"
+    o3, _ = ext().attention_fwd(qkv, seqlens, NH, p, 99, 2)
+    assert not torch.equal(o1, o3), "different offset must differ"
+    # dropped-mean check: E[out_p] == out_0 within tolerance
+    o0, _ = ext().attention_fwd(qkv, seqlens, NH, 0.0, 0, 0)
+    acc = torch.zeros_like(o0, dtype=torch.float32)
+    n = 32
+    for i in range(n):
+        oi, _ = ext().attention_fwd(qkv, seqlens, NH, p, 1234, 100 + i * 10**7)
+        acc += oi.float()
+    mean = acc / n
+    err = (mean - o0.float()).abs().mean() / o0.float().abs().mean()
+    assert err < 0.2, f"dropout mean deviates: {err}"
+    # backward runs and is finite
+    dq = ext().attention_bwd(
+        qkv[..., : H].contiguous(), qkv, seqlens, o1, l1, NH, p, 99, 1
+    )
+    assert torch.isfinite(dq.float()).all()
+
+
+# ---------------------------------------------------------------------------
+# multi-tensor optimizers
+# ---------------------------------------------------------------------------
+def _rand_tensors(seed, shapes):
+    g = torch.Generator(device=DEV).manual_seed(seed)
+    return [torch.randn(s, generator=g, device=DEV) for s in shapes]
+
+
+def test_l2norm_and_clip():
+    shapes = [(1000,), (37, 55), (256, 256), (3,)]
+    grads = _rand_tensors(11, shapes)
+    expected = torch.sqrt(sum(g.pow(2).sum() for g in grads))
+    gsq = ext().multi_tensor_l2norm_sq(grads)
+    assert abs(float(gsq.sqrt()) - float(expected)) / float(expected) < 1e-5
+    ext().multi_tensor_clip_scale(grads, gsq, 1.0)
+    after = torch.sqrt(sum(g.pow(2).sum() for g in grads))
+    assert abs(float(after) - 1.0) < 1e-4
+
+
+def test_fused_lamb_matches_eager():
+    from bert_pytorch_amd.optim import FusedLAMB
+
+    shapes = [(128, 128), (1024,), (30528, 8), (7,)]
+    torch.manual_seed(12)
+    params_g = [torch.randn(s, device=DEV) for s in shapes]
+    params_c = [p.clone() for p in params_g]
+    grads = [torch.randn(s, device=DEV) for s in shapes]
+
+    def build(params):
+        opt = FusedLAMB(
+            [
+                {"params": params[:2], "weight_decay": 0.01},
+                {"params": params[2:], "weight_decay": 0.0},
+            ],
+            lr=1e-2,
+        )
+        return opt
+
+    opt_native = build([p.requires_grad_(True) for p in params_g])
+    opt_eager = build([p.requires_grad_(True) for p in params_c])
+    for step in range(5):
+        for p, pc, g in zip(params_g, params_c, grads):
+            p.grad = (g * (step + 1)).clone()
+            pc.grad = (g * (step + 1)).clone()
+        opt_native.step()
+        os.environ["BPA_FORCE_EAGER"] = "1"
+        try:
+            opt_eager.step()
+        finally:
+            del os.environ["BPA_FORCE_EAGER"]
+    for p, pc in zip(params_g, params_c):
+        assert rel_err(p, pc) < 1e-4, "HIP LAMB diverges from eager LAMB"
+
+
+def test_fused_adam_matches_eager():
+    from bert_pytorch_amd.optim import FusedAdam
+
+    torch.manual_seed(13)
+    shapes = [(333,), (64, 64)]
+    params_g = [torch.randn(s, device=DEV).requires_grad_(True) for s in shapes]
+    params_c = [p.detach().clone().requires_grad_(True) for p in params_g]
+    opt_n = FusedAdam(params_g, lr=1e-2, weight_decay=0.01)
+    opt_e = FusedAdam(params_c, lr=1e-2, weight_decay=0.01)
+    for step in range(5):
+        g = [torch.randn(s, device=DEV) for s in shapes]
+        for p, pc, gi in zip(params_g, params_c, g):
+            p.grad = gi.clone()
+            pc.grad = gi.clone()
+        opt_n.step()
+        os.environ["BPA_FORCE_EAGER"] = "1"
+        try:
+            opt_e.step()
+        finally:
+            del os.environ["BPA_FORCE_EAGER"]
+    for p, pc in zip(params_g, params_c):
+        assert rel_err(p, pc) < 1e-4
+
+
+# ---------------------------------------------------------------------------
+# end-to-end on GPU
+# ---------------------------------------------------------------------------
+def test_model_step_bf16():
+    from bert_pytorch_amd.config import BertConfig
+    from bert_pytorch_amd.models import (
+        BertForPreTraining,
+        BertPretrainingCriterion,
+    )
+    from bert_pytorch_amd.optim import FusedLAMB
+
+    torch.manual_seed(14)
+    config = BertConfig(
+        vocab_size_or_config_json_file=2048, hidden_size=256,
+        num_hidden_layers=2, num_attention_heads=4, intermediate_size=512,
+        max_position_embeddings=128,
+    )
+    model = BertForPreTraining(config).to(DEV)
+    criterion = BertPretrainingCriterion(config.vocab_size)
+    opt = FusedLAMB(model.parameters(), lr=1e-3)
+    losses = []
+    ids = torch.randint(0, 2048, (8, 128), device=DEV)
+    tt = torch.zeros_like(ids)
+    mask = torch.ones_like(ids)
+    labels = torch.full_like(ids, -1)
+    labels[:, 4:12] = ids[:, 4:12]
+    nsp = torch.randint(0, 2, (8,), device=DEV)
+    for _ in range(8):
+        opt.zero_grad()
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            scores, rel, gl = model(ids, tt, mask, masked_lm_labels=labels)
+            loss = criterion(scores, rel, gl, nsp)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert all(l == l for l in losses), f"NaN loss: {losses}"
+    assert losses[-1] < losses[0], f"loss did not drop: {losses}"
+
+
+def test_gpu_matches_cpu_forward():
+    """Same weights/input: GPU HIP path vs CPU eager path agree."""
+    from bert_pytorch_amd.config import BertConfig
+    from bert_pytorch_amd.models import BertForPreTraining
+
+    torch.manual_seed(15)
+    config = BertConfig(
+        vocab_size_or_config_json_file=1024, hidden_size=128,
+        num_hidden_layers=2, num_attention_heads=2, intermediate_size=256,
+        max_position_embeddings=64,
+    )
+    model = BertForPreTraining(config).eval()
+    ids = torch.randint(0, 1024, (2, 64))
+    mask = torch.ones_like(ids)
+    mask[:, 50:] = 0
+    tt = torch.zeros_like(ids)
+    with torch.no_grad():
+        s_cpu, r_cpu = model(ids, tt, mask)
+    model_gpu = model.to(DEV)
+    with torch.no_grad(), torch.autocast("cuda", dtype=torch.bfloat16):
+        s_gpu, r_gpu = model_gpu(ids.to(DEV), tt.to(DEV), mask.to(DEV))
+    assert rel_err(s_gpu[:, :50], s_cpu[:, :50].to(DEV)) < 0.08
