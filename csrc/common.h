@@ -24,6 +24,9 @@ namespace bpa {
 template <typename T>
 __host__ __device__ __forceinline__ T tmin(T a, T b) { return a < b ? a : b; }
 
+// column-reduce partials chunk (col_reduce_kernel grid.y granularity)
+constexpr int kColChunk = 16;
+
 // ---------------------------------------------------------------------------
 // dtype conversion helpers (bf16/fp16/fp32 <-> fp32 compute)
 // ---------------------------------------------------------------------------

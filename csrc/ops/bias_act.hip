@@ -47,9 +47,8 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
   const int lane = threadIdx.x & (WAVE_SIZE - 1);
   const int wave = threadIdx.x / WAVE_SIZE;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* lb = reinterpret_cast<float*>(smem_raw);  // [H]
-  for (int c = threadIdx.x; c < H; c += blockDim.x) lb[c] = 0.f;
-  __syncthreads();
+  float* lb = reinterpret_cast<float*>(smem_raw) + wave * H;  // [NW][H]
+  for (int c = lane; c < H; c += WAVE_SIZE) lb[c] = 0.f;
 
   const int64_t row0 = static_cast<int64_t>(blockIdx.x) * rows_per_block;
   const int64_t row_end = min(row0 + rows_per_block, rows);
@@ -66,14 +65,20 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
         float g = DTraits<T>::to_f32(dv[k]) *
                   gelu_bwd(DTraits<T>::to_f32(xv[k]) + bias[c + k]);
         o[k] = DTraits<T>::from_f32(g);
-        atomicAdd(&lb[c + k], g);
+        lb[c + k] += g;  // lane-owned columns of the wave's slab
       }
       *reinterpret_cast<uint4*>(dxr + c) = *reinterpret_cast<const uint4*>(o);
     }
   }
   __syncthreads();
+  float* slab0 = reinterpret_cast<float*>(smem_raw);
   float* pb = part_dbias + static_cast<int64_t>(blockIdx.x) * H;
-  for (int c = threadIdx.x; c < H; c += blockDim.x) pb[c] = lb[c];
+  for (int c = threadIdx.x; c < H; c += blockDim.x) {
+    float acc = 0.f;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) acc += slab0[w * H + c];
+    pb[c] = acc;
+  }
 }
 
 #define DISPATCH_FLOATING2(TYPE, NAME, ...)                                  \
@@ -127,13 +132,18 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
   const int nblocks = static_cast<int>((rows + rows_per_block - 1) / rows_per_block);
   auto opts = x.options().dtype(torch::kFloat32);
   auto part_b = torch::empty({nblocks, H}, opts);
-  auto dbias = torch::empty({H}, opts);
+  auto dbias = torch::zeros({H}, opts);
   auto stream = at::hip::getCurrentHIPStream();
-  const size_t lds = static_cast<size_t>(H) * sizeof(float);
+  const size_t lds = NW * static_cast<size_t>(H) * sizeof(float);
   TORCH_CHECK(lds <= 160 * 1024, "bias_gelu_bwd: H too large");
   auto dy_c = dy.contiguous();
   DISPATCH_FLOATING2(x.scalar_type(), "bias_gelu_bwd", [&] {
     TORCH_CHECK(H % kVec == 0, "bias_gelu_bwd: H % ", kVec, " != 0");
+    if (lds > 48 * 1024) {
+      HIP_CHECK(hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&bias_gelu_bwd_kernel<scalar_t, kVec, NW>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, lds));
+    }
     hipLaunchKernelGGL((bias_gelu_bwd_kernel<scalar_t, kVec, NW>),
                        dim3(nblocks), dim3(NW * WAVE_SIZE), lds, stream,
                        reinterpret_cast<const scalar_t*>(dy_c.data_ptr()),
@@ -142,7 +152,7 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
                        reinterpret_cast<scalar_t*>(dx.data_ptr()),
                        part_b.data_ptr<float>(), rows, H, rows_per_block);
   });
-  dim3 rgrid((H + 255) / 256), rblock(256);
+  dim3 rgrid((H + 255) / 256, (nblocks + kColChunk - 1) / kColChunk), rblock(256);
   hipLaunchKernelGGL(col_reduce_kernel, rgrid, rblock, 0, stream,
                      part_b.data_ptr<float>(), nblocks, H,
                      dbias.data_ptr<float>());

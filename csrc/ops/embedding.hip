@@ -103,10 +103,9 @@ __global__ void embed_bwd_dz_kernel(
   const int lane = threadIdx.x & (WAVE_SIZE - 1);
   const int wave = threadIdx.x / WAVE_SIZE;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* lg = reinterpret_cast<float*>(smem_raw);
+  float* lg = reinterpret_cast<float*>(smem_raw) + wave * 2 * H;  // [NW][2H]
   float* lb = lg + H;
-  for (int c = threadIdx.x; c < 2 * H; c += blockDim.x) lg[c] = 0.f;
-  __syncthreads();
+  for (int c = lane; c < 2 * H; c += WAVE_SIZE) lg[c] = 0.f;
   const float keep_scale = TRAIN_DROP ? 1.0f / (1.0f - p) : 1.0f;
 
   const int row0 = blockIdx.x * rows_per_block;
@@ -135,8 +134,8 @@ __global__ void embed_bwd_dz_kernel(
         float dw = d * gamma[c + k];
         s1 += dw * zh;
         s2 += dw;
-        atomicAdd(&lg[c + k], d * zh);
-        atomicAdd(&lb[c + k], d);
+        lg[c + k] += d * zh;
+        lb[c + k] += d;
         dz[base + c + k] = dw;  // stash dw; finalized below after row sums
       }
     }
@@ -151,9 +150,16 @@ __global__ void embed_bwd_dz_kernel(
     }
   }
   __syncthreads();
+  float* slab0 = reinterpret_cast<float*>(smem_raw);
   for (int c = threadIdx.x; c < H; c += blockDim.x) {
-    part_dgamma[static_cast<int64_t>(blockIdx.x) * H + c] = lg[c];
-    part_dbeta[static_cast<int64_t>(blockIdx.x) * H + c] = lb[c];
+    float ag = 0.f, ab = 0.f;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) {
+      ag += slab0[w * 2 * H + c];
+      ab += slab0[w * 2 * H + H + c];
+    }
+    part_dgamma[static_cast<int64_t>(blockIdx.x) * H + c] = ag;
+    part_dbeta[static_cast<int64_t>(blockIdx.x) * H + c] = ab;
   }
 }
 
@@ -309,9 +315,16 @@ std::vector<torch::Tensor> embedding_ln_dropout_bwd(
   auto part_g = torch::empty({nblocks, H}, fopts);
   auto part_b = torch::empty({nblocks, H}, fopts);
   auto stream = at::hip::getCurrentHIPStream();
-  const size_t lds = 2 * static_cast<size_t>(H) * sizeof(float);
+  const size_t lds = NW * 2 * static_cast<size_t>(H) * sizeof(float);
   DISPATCH_OUT(dy.scalar_type(), "embed_bwd", [&] {
     auto launch = [&](auto train_c) {
+      if (lds > 48 * 1024) {
+        HIP_CHECK(hipFuncSetAttribute(
+            reinterpret_cast<const void*>(
+                &embed_bwd_dz_kernel<out_t, kVec, NW,
+                                     decltype(train_c)::value>),
+            hipFuncAttributeMaxDynamicSharedMemorySize, lds));
+      }
       hipLaunchKernelGGL(
           (embed_bwd_dz_kernel<out_t, kVec, NW, decltype(train_c)::value>),
           dim3(nblocks), dim3(NW * WAVE_SIZE), lds, stream,
@@ -330,10 +343,10 @@ std::vector<torch::Tensor> embedding_ln_dropout_bwd(
   auto d_pos = torch::zeros({max_pos, H}, fopts);
   auto d_tok = has_tok ? torch::zeros({n_types, H}, fopts)
                        : torch::empty({0}, fopts);
-  auto dgamma = torch::empty({H}, fopts);
-  auto dbeta = torch::empty({H}, fopts);
+  auto dgamma = torch::zeros({H}, fopts);
+  auto dbeta = torch::zeros({H}, fopts);
 
-  dim3 rgrid((H + 255) / 256), rblock(256);
+  dim3 rgrid((H + 255) / 256, (nblocks + kColChunk - 1) / kColChunk), rblock(256);
   hipLaunchKernelGGL(col_reduce_kernel, rgrid, rblock, 0, stream,
                      part_g.data_ptr<float>(), nblocks, H,
                      dgamma.data_ptr<float>());
@@ -360,7 +373,9 @@ std::vector<torch::Tensor> embedding_ln_dropout_bwd(
                        dz.data_ptr<float>(), tt_c.data_ptr<int64_t>(),
                        tparts.data_ptr<float>(), rows, H,
                        static_cast<int>(n_types), rows_per_block);
-    dim3 tgrid((n_types * H + 255) / 256);
+    dim3 tgrid((n_types * H + 255) / 256,
+               (nblocks + kColChunk - 1) / kColChunk);
+    d_tok.zero_();
     hipLaunchKernelGGL(col_reduce_kernel, tgrid, rblock, 0, stream,
                        tparts.data_ptr<float>(), nblocks,
                        static_cast<int>(n_types * H),

@@ -103,12 +103,12 @@ __global__ void bdrl_bwd_kernel(
   const int lane = threadIdx.x & (WAVE_SIZE - 1);
   const int wave = threadIdx.x / WAVE_SIZE;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* lg = reinterpret_cast<float*>(smem_raw);  // [H] dgamma
-  float* lb = lg + H;                              // [H] dbeta
-  float* lbias = lb + H;                           // [H] dbias
   const int n_slabs = HAS_BIAS ? 3 : 2;
-  for (int c = threadIdx.x; c < n_slabs * H; c += blockDim.x) lg[c] = 0.f;
-  __syncthreads();
+  // per-wave private slabs [NW][n_slabs][H]: no atomics, no contention
+  float* lg = reinterpret_cast<float*>(smem_raw) + wave * n_slabs * H;
+  float* lb = lg + H;
+  float* lbias = lb + H;
+  for (int c = lane; c < n_slabs * H; c += WAVE_SIZE) lg[c] = 0.f;
   const float keep_scale = TRAIN_DROP ? 1.0f / (1.0f - p) : 1.0f;
 
   const int row0 = blockIdx.x * rows_per_block;
@@ -128,8 +128,8 @@ __global__ void bdrl_bwd_kernel(
         float dw = d * gamma[c + k];
         s1 += dw * zh;
         s2 += dw;
-        atomicAdd(&lg[c + k], d * zh);
-        atomicAdd(&lb[c + k], d);
+        lg[c + k] += d * zh;
+        lb[c + k] += d;
       }
     }
     s1 = wave_reduce_sum(s1) / H;
@@ -156,7 +156,7 @@ __global__ void bdrl_bwd_kernel(
         dzo[k] = DTraits<T>::from_f32(dzk);  // grad to residual input
         float dxk = TRAIN_DROP ? (mv[k] ? dzk * keep_scale : 0.f) : dzk;
         dxo[k] = DTraits<T>::from_f32(dxk);
-        if (HAS_BIAS) atomicAdd(&lbias[c + k], dxk);
+        if (HAS_BIAS) lbias[c + k] += dxk;
       }
       *reinterpret_cast<uint4*>(dz_res + base + c) =
           *reinterpret_cast<const uint4*>(dzo);
@@ -165,11 +165,19 @@ __global__ void bdrl_bwd_kernel(
     }
   }
   __syncthreads();
+  float* slab0 = reinterpret_cast<float*>(smem_raw);
   for (int c = threadIdx.x; c < H; c += blockDim.x) {
-    part_dgamma[static_cast<int64_t>(blockIdx.x) * H + c] = lg[c];
-    part_dbeta[static_cast<int64_t>(blockIdx.x) * H + c] = lb[c];
+    float ag = 0.f, ab = 0.f, abias = 0.f;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) {
+      ag += slab0[w * n_slabs * H + c];
+      ab += slab0[w * n_slabs * H + H + c];
+      if (HAS_BIAS) abias += slab0[w * n_slabs * H + 2 * H + c];
+    }
+    part_dgamma[static_cast<int64_t>(blockIdx.x) * H + c] = ag;
+    part_dbeta[static_cast<int64_t>(blockIdx.x) * H + c] = ab;
     if (HAS_BIAS)
-      part_dbias[static_cast<int64_t>(blockIdx.x) * H + c] = lbias[c];
+      part_dbias[static_cast<int64_t>(blockIdx.x) * H + c] = abias;
   }
 }
 
@@ -267,11 +275,20 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
       has_bias ? torch::empty({nblocks, H}, fopts) : torch::empty({0}, fopts);
   const bool train_drop = p > 0.0;
   auto stream = at::hip::getCurrentHIPStream();
-  const size_t lds = (has_bias ? 3 : 2) * static_cast<size_t>(H) * sizeof(float);
+  const size_t lds =
+      NW * (has_bias ? 3 : 2) * static_cast<size_t>(H) * sizeof(float);
   TORCH_CHECK(lds <= 160 * 1024, "bdrl_bwd: H too large");
   DISPATCH_T(z.scalar_type(), "bdrl_bwd", [&] {
     TORCH_CHECK(H % kVec == 0, "bdrl_bwd: H % ", kVec, " != 0");
     auto launch = [&](auto has_bias_c, auto train_c) {
+      if (lds > 48 * 1024) {
+        HIP_CHECK(hipFuncSetAttribute(
+            reinterpret_cast<const void*>(
+                &bdrl_bwd_kernel<scalar_t, kVec, NW,
+                                 decltype(has_bias_c)::value,
+                                 decltype(train_c)::value>),
+            hipFuncAttributeMaxDynamicSharedMemorySize, lds));
+      }
       hipLaunchKernelGGL(
           (bdrl_bwd_kernel<scalar_t, kVec, NW, decltype(has_bias_c)::value,
                            decltype(train_c)::value>),
@@ -296,10 +313,10 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
     else
       launch(std::false_type{}, std::false_type{});
   });
-  auto dgamma = torch::empty({H}, fopts);
-  auto dbeta = torch::empty({H}, fopts);
-  auto dbias = has_bias ? torch::empty({H}, fopts) : torch::empty({0}, fopts);
-  dim3 rgrid((H + 255) / 256), rblock(256);
+  auto dgamma = torch::zeros({H}, fopts);
+  auto dbeta = torch::zeros({H}, fopts);
+  auto dbias = has_bias ? torch::zeros({H}, fopts) : torch::empty({0}, fopts);
+  dim3 rgrid((H + 255) / 256, (nblocks + kColChunk - 1) / kColChunk), rblock(256);
   hipLaunchKernelGGL(col_reduce_kernel, rgrid, rblock, 0, stream,
                      part_g.data_ptr<float>(), nblocks, H,
                      dgamma.data_ptr<float>());

@@ -76,10 +76,10 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
   const int lane = threadIdx.x & (WAVE_SIZE - 1);
   const int wave = threadIdx.x / WAVE_SIZE;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* lg = reinterpret_cast<float*>(smem_raw);      // [H]
-  float* lb = lg + H;                                  // [H]
-  for (int c = threadIdx.x; c < 2 * H; c += blockDim.x) lg[c] = 0.f;
-  __syncthreads();
+  // per-wave private slabs: no atomics, no cross-wave contention
+  float* lg = reinterpret_cast<float*>(smem_raw) + wave * 2 * H;  // [NW][2H]
+  float* lb = lg + H;
+  for (int c = lane; c < 2 * H; c += WAVE_SIZE) lg[c] = 0.f;
 
   const int row0 = blockIdx.x * rows_per_block;
   const int row_end = min(row0 + rows_per_block, rows);
@@ -116,28 +116,39 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
         float xh = (DTraits<T>::to_f32(xv[k]) - mu) * rs;
         float dw = d * gamma[c + k];
         o[k] = DTraits<T>::from_f32(rs * (dw - s2 - xh * s1));
-        atomicAdd(&lg[c + k], d * xh);  // LDS atomic (ds_add)
-        atomicAdd(&lb[c + k], d);
+        lg[c + k] += d * xh;  // this lane owns these columns in its slab
+        lb[c + k] += d;
       }
       *reinterpret_cast<uint4*>(dxr + c) = *reinterpret_cast<const uint4*>(o);
     }
   }
   __syncthreads();
+  float* slab0 = reinterpret_cast<float*>(smem_raw);
   float* pg = part_dgamma + static_cast<int64_t>(blockIdx.x) * H;
   float* pb = part_dbeta + static_cast<int64_t>(blockIdx.x) * H;
   for (int c = threadIdx.x; c < H; c += blockDim.x) {
-    pg[c] = lg[c];
-    pb[c] = lb[c];
+    float ag = 0.f, ab = 0.f;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) {
+      ag += slab0[w * 2 * H + c];
+      ab += slab0[w * 2 * H + H + c];
+    }
+    pg[c] = ag;
+    pb[c] = ab;
   }
 }
 
+// grid (ceil(H/256), ceil(nparts/kColChunk)); out must be zero-filled.
 __global__ void col_reduce_kernel(const float* __restrict__ parts, int nparts,
                                   int H, float* __restrict__ out) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= H) return;
+  const int p0 = blockIdx.y * kColChunk;
+  const int p1 = min(p0 + kColChunk, nparts);
   float acc = 0.f;
-  for (int p = 0; p < nparts; ++p) acc += parts[static_cast<int64_t>(p) * H + c];
-  out[c] = acc;
+  for (int p = p0; p < p1; ++p)
+    acc += parts[static_cast<int64_t>(p) * H + c];
+  atomicAdd(&out[c], acc);
 }
 
 #define DISPATCH_FLOATING(TYPE, NAME, ...)                                   \
@@ -197,11 +208,11 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x,
   auto opts = x.options().dtype(torch::kFloat32);
   auto part_g = torch::empty({nblocks, H}, opts);
   auto part_b = torch::empty({nblocks, H}, opts);
-  auto dgamma = torch::empty({H}, opts);
-  auto dbeta = torch::empty({H}, opts);
+  auto dgamma = torch::zeros({H}, opts);
+  auto dbeta = torch::zeros({H}, opts);
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid(nblocks), block(NW * WAVE_SIZE);
-  const size_t lds = 2 * static_cast<size_t>(H) * sizeof(float);
+  const size_t lds = NW * 2 * static_cast<size_t>(H) * sizeof(float);
   TORCH_CHECK(lds <= 160 * 1024, "ln_bwd: H too large for LDS accumulation");
   auto dy_c = dy.contiguous();
   DISPATCH_FLOATING(x.scalar_type(), "ln_bwd", [&] {
@@ -216,7 +227,7 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x,
                        part_g.data_ptr<float>(), part_b.data_ptr<float>(), rows,
                        H, rows_per_block);
   });
-  dim3 rgrid((H + 255) / 256), rblock(256);
+  dim3 rgrid((H + 255) / 256, (nblocks + kColChunk - 1) / kColChunk), rblock(256);
   hipLaunchKernelGGL(col_reduce_kernel, rgrid, rblock, 0, stream,
                      part_g.data_ptr<float>(), nblocks, H,
                      dgamma.data_ptr<float>());
