@@ -79,6 +79,9 @@ def make_batch(gen, device, bsz, seq, vocab, max_pred):
 
 def main():
     args = parse_args()
+    if torch.cuda.is_available():
+        from bert_pytorch_amd.utils import tunable
+        tunable.enable()
     rank, local_rank, world = comm.init_distributed()
     if args.gpus > 1:
         assert world == args.gpus, f"WORLD_SIZE {world} != --gpus {args.gpus}"

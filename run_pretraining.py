@@ -91,6 +91,10 @@ def parse_arguments(args=None) -> argparse.Namespace:
 
 
 def setup_training(args):
+    if torch.cuda.is_available():
+        from bert_pytorch_amd.utils import tunable  # noqa: PLC0415
+
+        tunable.enable()
     rank, local_rank, world_size = comm.init_distributed()
     args.local_rank = local_rank
     device = (
