@@ -58,6 +58,19 @@ void fused_adam(std::vector<torch::Tensor> params,
 
 }  // namespace bpa
 
+namespace bpa_tok {
+int64_t create_wordpiece(std::vector<std::string> vocab, std::string unk);
+std::pair<std::vector<std::string>, std::vector<int64_t>> encode_wordpiece(
+    int64_t handle, std::vector<std::string> words);
+int64_t create_bpe(std::vector<std::string> vocab,
+                   std::vector<std::string> merge_lines);
+std::pair<std::vector<std::string>, std::vector<int64_t>> encode_bpe(
+    int64_t handle, std::vector<std::string> pretokens);
+std::vector<std::string> train_bpe(std::vector<std::string> words,
+                                   std::vector<int64_t> counts,
+                                   int64_t num_merges);
+}  // namespace bpa_tok
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "bert_pytorch_amd gfx950 HIP kernels";
   m.def("ln_fwd", &bpa::ln_fwd, "fused LayerNorm forward");
@@ -76,4 +89,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_clip_scale", &bpa::multi_tensor_clip_scale);
   m.def("fused_lamb", &bpa::fused_lamb);
   m.def("fused_adam", &bpa::fused_adam);
+  m.def("tok_create_wordpiece", &bpa_tok::create_wordpiece);
+  m.def("tok_encode_wordpiece", &bpa_tok::encode_wordpiece);
+  m.def("tok_create_bpe", &bpa_tok::create_bpe);
+  m.def("tok_encode_bpe", &bpa_tok::encode_bpe);
+  m.def("tok_train_bpe", &bpa_tok::train_bpe);
 }

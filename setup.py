@@ -23,6 +23,7 @@ sources = [
     "csrc/ops/cross_entropy.hip",
     "csrc/ops/attention.hip",
     "csrc/optim/multi_tensor.hip",
+    "csrc/tok/tokenizer.cpp",
 ]
 
 setup(
