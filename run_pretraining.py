@@ -186,6 +186,7 @@ def prepare_optimizers(args, model, resume_state):
 
         preconditioner = KFAC(
             model,
+            optimizer=optimizer,
             factor_update_interval=args.kfac_factor_interval,
             inv_update_interval=args.kfac_inv_interval,
             skip_layers=args.kfac_skip_layers,
