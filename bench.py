@@ -132,7 +132,7 @@ def main():
         if sync:
             scheduler.step()
             optimizer.step()
-            optimizer.zero_grad(set_to_none=False)
+            optimizer.zero_grad(set_to_none=True)
 
     model.train()
     for i in range(args.warmup):

@@ -2,9 +2,16 @@
 //
 // The bias+activation fusion point of the reference's LinearActivation
 // (src/modeling.py:141-185, jit bias_gelu at :126-139). Elementwise,
-// memory-bound: grid-stride waves, 16 B/lane vector IO. Backward also
-// produces dbias = column-sum(dx) via per-block LDS accumulation + a
-// deterministic column-reduce pass.
+// memory-bound: 2D grid (column-block, row) with no per-element integer
+// division, 16 B/lane vector IO, enough resident waves to hide HBM3E
+// latency (grid is NOT capped — [12288,4096] launches ~24k workgroups
+// across the 256 CUs / 8 XCDs).
+//
+// Backward fuses the dbias column reduction into the dx pass: each wave
+// keeps a per-lane fp32 accumulator over its row slice (registers, no
+// LDS traffic in the hot loop), waves combine through an 8 KiB LDS slab,
+// blocks write [grid.y, H] partials reduced by col_reduce_kernel. This
+// removes the old second full read of dx (100 MB at phase-1 FFN shapes).
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -20,60 +27,90 @@ __global__ void col_reduce_kernel(const float* __restrict__ parts, int nparts,
 template <typename T, int VEC>
 __global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
                                      const float* __restrict__ bias,
-                                     T* __restrict__ y, int64_t rows, int H) {
-  const int64_t total_vec = rows * (H / VEC);
-  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total_vec;
-       i += static_cast<int64_t>(gridDim.x) * blockDim.x) {
-    const int64_t e = i * VEC;
-    const int c = static_cast<int>(e % H);
+                                     T* __restrict__ y, int rows, int H) {
+  const int vcol = blockIdx.x * blockDim.x + threadIdx.x;
+  const int c = vcol * VEC;
+  if (c >= H) return;
+  float bv[VEC];
+#pragma unroll
+  for (int k = 0; k < VEC; ++k) bv[k] = bias[c + k];
+  for (int r = blockIdx.y; r < rows; r += gridDim.y) {
+    const int64_t e = static_cast<int64_t>(r) * H + c;
     T v[VEC], o[VEC];
     *reinterpret_cast<uint4*>(v) = *reinterpret_cast<const uint4*>(x + e);
 #pragma unroll
     for (int k = 0; k < VEC; ++k) {
-      o[k] = DTraits<T>::from_f32(
-          gelu_fwd(DTraits<T>::to_f32(v[k]) + bias[c + k]));
+      o[k] = DTraits<T>::from_f32(gelu_fwd(DTraits<T>::to_f32(v[k]) + bv[k]));
     }
     *reinterpret_cast<uint4*>(y + e) = *reinterpret_cast<const uint4*>(o);
   }
 }
 
-template <typename T, int VEC>
+// Backward: dx = dy * gelu'(x + bias); per-column dbias partials folded in.
+// Block: NW waves; wave w covers rows {r0+w, r0+w+NW, ...} of a
+// ROWS_PER_BLOCK row stripe and a 256*VEC-wide column window; per-lane
+// fp32 accumulators combine across waves through LDS at stripe end.
+template <typename T, int VEC, int NW>
 __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
                                      const T* __restrict__ x,
                                      const float* __restrict__ bias,
-                                     T* __restrict__ dx, int64_t rows, int H) {
-  const int64_t total_vec = rows * (H / VEC);
-  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total_vec;
-       i += static_cast<int64_t>(gridDim.x) * blockDim.x) {
-    const int64_t e = i * VEC;
-    const int c = static_cast<int>(e % H);
-    T dv[VEC], xv[VEC], o[VEC];
-    *reinterpret_cast<uint4*>(dv) = *reinterpret_cast<const uint4*>(dy + e);
-    *reinterpret_cast<uint4*>(xv) = *reinterpret_cast<const uint4*>(x + e);
-#pragma unroll
-    for (int k = 0; k < VEC; ++k) {
-      o[k] = DTraits<T>::from_f32(
-          DTraits<T>::to_f32(dv[k]) *
-          gelu_bwd(DTraits<T>::to_f32(xv[k]) + bias[c + k]));
-    }
-    *reinterpret_cast<uint4*>(dx + e) = *reinterpret_cast<const uint4*>(o);
-  }
-}
+                                     T* __restrict__ dx,
+                                     float* __restrict__ part_dbias, int rows,
+                                     int H, int rows_per_block) {
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const int wave = threadIdx.x / WAVE_SIZE;
+  // every wave covers the same 64-lane-wide column window (so the LDS
+  // combine lines up lane-for-lane) and a disjoint subset of the rows
+  const int c = (blockIdx.x * WAVE_SIZE + lane) * VEC;
 
-// column sum of a [rows, H] tensor into fp32 out[H] (atomic per chunk).
-// grid (ceil(H/256), ceil(rows/kColSumRows)); out must be zero-filled.
-constexpr int kColSumRows = 64;
-template <typename T>
-__global__ void col_sum_kernel(const T* __restrict__ src, int64_t rows, int H,
-                               float* __restrict__ out) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= H) return;
-  const int64_t r0 = static_cast<int64_t>(blockIdx.y) * kColSumRows;
-  const int64_t r1 = tmin<int64_t>(r0 + kColSumRows, rows);
-  float acc = 0.f;
-  for (int64_t r = r0; r < r1; ++r)
-    acc += DTraits<T>::to_f32(src[r * H + c]);
-  atomicAdd(&out[c], acc);
+  float acc[VEC];
+#pragma unroll
+  for (int k = 0; k < VEC; ++k) acc[k] = 0.f;
+  float bv[VEC];
+  if (c < H) {
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) bv[k] = bias[c + k];
+  }
+
+  const int r0 = blockIdx.y * rows_per_block;
+  const int r1 = min(r0 + rows_per_block, rows);
+  if (c < H) {
+    for (int r = r0 + wave; r < r1; r += NW) {
+      const int64_t e = static_cast<int64_t>(r) * H + c;
+      T dv[VEC], xv[VEC], o[VEC];
+      *reinterpret_cast<uint4*>(dv) = *reinterpret_cast<const uint4*>(dy + e);
+      *reinterpret_cast<uint4*>(xv) = *reinterpret_cast<const uint4*>(x + e);
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        const float g = DTraits<T>::to_f32(dv[k]) *
+                        gelu_bwd(DTraits<T>::to_f32(xv[k]) + bv[k]);
+        o[k] = DTraits<T>::from_f32(g);
+        acc[k] += g;
+      }
+      *reinterpret_cast<uint4*>(dx + e) = *reinterpret_cast<const uint4*>(o);
+    }
+  }
+
+  // cross-wave combine: [NW][64*VEC] fp32 slab (NW=4, VEC=8 -> 8 KiB)
+  extern __shared__ __attribute__((aligned(16))) float slab[];
+  float* mine = slab + (wave * WAVE_SIZE + lane) * VEC;
+#pragma unroll
+  for (int k = 0; k < VEC; ++k) mine[k] = acc[k];
+  __syncthreads();
+  if (wave == 0 && c < H) {
+    float tot[VEC];
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) tot[k] = 0.f;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) {
+      const float* src = slab + (w * WAVE_SIZE + lane) * VEC;
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) tot[k] += src[k];
+    }
+    float* dst = part_dbias + static_cast<int64_t>(blockIdx.y) * H + c;
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) dst[k] = tot[k];
+  }
 }
 
 #define DISPATCH_FLOATING2(TYPE, NAME, ...)                                  \
@@ -104,14 +141,16 @@ torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias) {
   auto stream = at::hip::getCurrentHIPStream();
   DISPATCH_FLOATING2(x.scalar_type(), "bias_gelu_fwd", [&] {
     TORCH_CHECK(H % kVec == 0, "bias_gelu_fwd: H % ", kVec, " != 0");
-    const int64_t total = rows * (H / kVec);
-    const int blocks =
-        static_cast<int>(tmin<int64_t>((total + 255) / 256, 2048));
-    hipLaunchKernelGGL((bias_gelu_fwd_kernel<scalar_t, kVec>), dim3(blocks),
-                       dim3(256), 0, stream,
+    const int vec_per_row = H / kVec;
+    const int threads = tmin(256, ((vec_per_row + 63) / 64) * 64);
+    dim3 grid((vec_per_row + threads - 1) / threads,
+              static_cast<unsigned>(tmin<int64_t>(rows, 65535)));
+    hipLaunchKernelGGL((bias_gelu_fwd_kernel<scalar_t, kVec>), grid,
+                       dim3(threads), 0, stream,
                        reinterpret_cast<const scalar_t*>(x.data_ptr()),
                        bias_f.data_ptr<float>(),
-                       reinterpret_cast<scalar_t*>(y.data_ptr()), rows, H);
+                       reinterpret_cast<scalar_t*>(y.data_ptr()),
+                       static_cast<int>(rows), H);
   });
   return y;
 }
@@ -122,27 +161,33 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
   const int H = x.size(1);
   auto bias_f = bias.contiguous().to(torch::kFloat32);
   auto dx = torch::empty_like(x);
-  auto opts = x.options().dtype(torch::kFloat32);
-  auto dbias = torch::zeros({H}, opts);
+  auto fopts = x.options().dtype(torch::kFloat32);
   auto stream = at::hip::getCurrentHIPStream();
+  constexpr int NW = 4;
+  const int rows_per_block = 16;
+  const int stripe_count =
+      static_cast<int>((rows + rows_per_block - 1) / rows_per_block);
+  auto dbias = torch::zeros({H}, fopts);
   auto dy_c = dy.contiguous();
   DISPATCH_FLOATING2(x.scalar_type(), "bias_gelu_bwd", [&] {
     TORCH_CHECK(H % kVec == 0, "bias_gelu_bwd: H % ", kVec, " != 0");
-    const int64_t total = rows * (H / kVec);
-    const int blocks =
-        static_cast<int>(tmin<int64_t>((total + 255) / 256, 2048));
-    hipLaunchKernelGGL((bias_gelu_bwd_kernel<scalar_t, kVec>), dim3(blocks),
-                       dim3(256), 0, stream,
+    auto part = torch::empty({stripe_count, H}, fopts);
+    const int lanes_per_row = WAVE_SIZE;  // one wave per row slice
+    dim3 grid((H / kVec + lanes_per_row - 1) / lanes_per_row, stripe_count);
+    const size_t lds = NW * WAVE_SIZE * kVec * sizeof(float);
+    hipLaunchKernelGGL((bias_gelu_bwd_kernel<scalar_t, kVec, NW>), grid,
+                       dim3(NW * WAVE_SIZE), lds, stream,
                        reinterpret_cast<const scalar_t*>(dy_c.data_ptr()),
                        reinterpret_cast<const scalar_t*>(x.data_ptr()),
                        bias_f.data_ptr<float>(),
-                       reinterpret_cast<scalar_t*>(dx.data_ptr()), rows, H);
-    dim3 sgrid((H + 255) / 256,
-               static_cast<unsigned>((rows + kColSumRows - 1) / kColSumRows));
-    hipLaunchKernelGGL((col_sum_kernel<scalar_t>), sgrid, dim3(256), 0,
-                       stream,
-                       reinterpret_cast<const scalar_t*>(dx.data_ptr()), rows,
-                       H, dbias.data_ptr<float>());
+                       reinterpret_cast<scalar_t*>(dx.data_ptr()),
+                       part.data_ptr<float>(), static_cast<int>(rows), H,
+                       rows_per_block);
+    dim3 rgrid((H + 255) / 256,
+               (stripe_count + kColChunk - 1) / kColChunk);
+    hipLaunchKernelGGL(col_reduce_kernel, rgrid, dim3(256), 0, stream,
+                       part.data_ptr<float>(), stripe_count, H,
+                       dbias.data_ptr<float>());
   });
   if (bias.scalar_type() != torch::kFloat32) {
     return {dx, dbias.to(bias.scalar_type())};

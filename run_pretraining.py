@@ -18,6 +18,7 @@ Launch (single node, one process per GPU):
 from __future__ import annotations
 
 import argparse
+import contextlib
 import math
 import os
 import random
@@ -33,6 +34,7 @@ from bert_pytorch_amd.models import BertForPreTraining, BertPretrainingCriterion
 from bert_pytorch_amd.optim import FusedLAMB, PolyWarmUpScheduler
 from bert_pytorch_amd.parallel import comm
 from bert_pytorch_amd.utils import MetricLogger, checkpoint as ckpt_io
+from bert_pytorch_amd.utils.profiling import StepTimer, torch_profile
 
 MASK_TOKEN_DEFAULT = 103  # BERT [MASK]; overridden by --vocab_file when given
 
@@ -79,6 +81,11 @@ def parse_arguments(args=None) -> argparse.Namespace:
     parser.add_argument("--kfac_factor_interval", type=int, default=1)
     parser.add_argument("--kfac_skip_layers", nargs="+",
                         default=["BertLMPredictionHead", "embedding"])
+    parser.add_argument("--profile", type=int, default=0,
+                        help="profile N optimizer steps with torch.profiler "
+                             "(chrome trace under output_dir/profile) and stop")
+    parser.add_argument("--timing_breakdown", action="store_true",
+                        help="log per-step data/h2d/fwd/bwd/opt breakdown")
     parser.add_argument("--disable_progress_bar", action="store_true")
     parser.add_argument("--local_rank", type=int,
                         default=int(os.environ.get("LOCAL_RANK", 0)))
@@ -272,10 +279,10 @@ def prepare_dataset(args, resume_state):
 
 
 def forward_backward_pass(model, criterion, scaler, batch, args, sync_grads,
-                          autocast_dtype):
+                          autocast_dtype, timer):
     input_ids, segment_ids, input_mask, mlm_labels, nsp_labels = batch
     enabled = autocast_dtype is not None
-    with torch.autocast(
+    with timer.phase("forward"), torch.autocast(
         device_type="cuda" if input_ids.is_cuda else "cpu",
         dtype=autocast_dtype or torch.bfloat16,
         enabled=enabled,
@@ -285,13 +292,14 @@ def forward_backward_pass(model, criterion, scaler, batch, args, sync_grads,
         )
         loss = criterion(scores, seq_rel, gathered_labels, nsp_labels)
         loss = loss / args.accumulation_steps
-    if sync_grads or not isinstance(
-        model, torch.nn.parallel.DistributedDataParallel
-    ):
-        scaler.scale(loss).backward()
-    else:
-        with model.no_sync():
+    with timer.phase("backward"):
+        if sync_grads or not isinstance(
+            model, torch.nn.parallel.DistributedDataParallel
+        ):
             scaler.scale(loss).backward()
+        else:
+            with model.no_sync():
+                scaler.scale(loss).backward()
     return loss.detach()
 
 
@@ -303,7 +311,7 @@ def take_optimizer_step(optimizer, scheduler, scaler, model, preconditioner):
         preconditioner.step()
     scaler.step(optimizer)
     scaler.update()
-    optimizer.zero_grad(set_to_none=False)
+    optimizer.zero_grad(set_to_none=True)
 
 
 def main(args) -> int:
@@ -329,6 +337,17 @@ def main(args) -> int:
     steps_this_run = args.steps if args.steps > 0 else args.max_steps
     start_steps = global_steps
 
+    timer = StepTimer(enabled=args.timing_breakdown)
+    profile_stack = contextlib.ExitStack()
+    if args.profile > 0:
+        steps_this_run = min(steps_this_run, args.profile)
+        profile_stack.enter_context(
+            torch_profile(
+                os.path.join(args.output_dir or ".", "profile"),
+                rank=comm.get_rank(),
+            )
+        )
+
     micro_step = 0
     accum_loss = 0.0
     window_samples = 0
@@ -343,69 +362,85 @@ def main(args) -> int:
         args.global_batch_size, args.max_steps, steps_this_run, global_steps,
     )
 
-    while not done:
-        sampler.set_epoch(epoch)
-        for batch in loader:
-            micro_step += 1
-            batch = [t.to(device, non_blocking=True) for t in batch]
-            sync_grads = micro_step % args.accumulation_steps == 0
-            loss = forward_backward_pass(
-                model, criterion, scaler, batch, args, sync_grads,
-                autocast_dtype,
-            )
-            accum_loss += float(loss)
-            window_samples += batch[0].shape[0]
-            if not sync_grads:
-                continue
+    def timed_iter(data_loader):
+        it = iter(data_loader)
+        while True:
+            with timer.phase("data"):
+                try:
+                    item = next(it)
+                except StopIteration:
+                    return
+            yield item
 
-            take_optimizer_step(
-                optimizer, scheduler, scaler, model, preconditioner
-            )
-            global_steps += 1
-            if train_start is None:
-                train_start = time.perf_counter()  # skip step-0 warmup cost
-
-            if global_steps % 10 == 0 or global_steps <= 2:
-                now = time.perf_counter()
-                samples_per_second = (
-                    window_samples * comm.get_world_size() / (now - window_start)
+    with profile_stack:
+        while not done:
+            sampler.set_epoch(epoch)
+            for batch in timed_iter(loader):
+                micro_step += 1
+                with timer.phase("h2d"):
+                    batch = [t.to(device, non_blocking=True) for t in batch]
+                sync_grads = micro_step % args.accumulation_steps == 0
+                loss = forward_backward_pass(
+                    model, criterion, scaler, batch, args, sync_grads,
+                    autocast_dtype, timer,
                 )
-                if comm.is_main_process():
-                    log.log(
-                        "train",
-                        global_steps + args.previous_phase_end_step,
-                        epoch=epoch,
-                        average_loss=accum_loss,
-                        step_loss=float(loss) * args.accumulation_steps,
-                        learning_rate=optimizer.param_groups[0]["lr"],
-                        samples_per_second=samples_per_second,
+                accum_loss += float(loss)
+                window_samples += batch[0].shape[0]
+                if not sync_grads:
+                    continue
+
+                with timer.phase("optimizer"):
+                    take_optimizer_step(
+                        optimizer, scheduler, scaler, model, preconditioner
                     )
-                window_start = now
-                window_samples = 0
-            accum_loss = 0.0
+                timer.step_end()
+                global_steps += 1
+                if train_start is None:
+                    train_start = time.perf_counter()  # skip step-0 warmup cost
 
-            if (
-                args.output_dir
-                and global_steps % args.num_steps_per_checkpoint == 0
-                and comm.is_main_process()
-            ):
-                _save_checkpoint(
-                    args, model, optimizer, sampler, scaler, preconditioner,
-                    epoch, global_steps,
-                )
-            if (
-                global_steps >= args.max_steps
-                or global_steps - start_steps >= steps_this_run
-            ):
-                done = True
-                break
-        epoch += 1
+                if global_steps % 10 == 0 or global_steps <= 2:
+                    now = time.perf_counter()
+                    samples_per_second = (
+                        window_samples * comm.get_world_size() / (now - window_start)
+                    )
+                    if comm.is_main_process():
+                        log.log(
+                            "train",
+                            global_steps + args.previous_phase_end_step,
+                            epoch=epoch,
+                            average_loss=accum_loss,
+                            step_loss=float(loss) * args.accumulation_steps,
+                            learning_rate=optimizer.param_groups[0]["lr"],
+                            samples_per_second=samples_per_second,
+                        )
+                    window_start = now
+                    window_samples = 0
+                accum_loss = 0.0
+
+                if (
+                    args.output_dir
+                    and global_steps % args.num_steps_per_checkpoint == 0
+                    and comm.is_main_process()
+                ):
+                    _save_checkpoint(
+                        args, model, optimizer, sampler, scaler, preconditioner,
+                        epoch, global_steps,
+                    )
+                if (
+                    global_steps >= args.max_steps
+                    or global_steps - start_steps >= steps_this_run
+                ):
+                    done = True
+                    break
+            epoch += 1
 
     if args.output_dir and comm.is_main_process():
         _save_checkpoint(
             args, model, optimizer, sampler, scaler, preconditioner, epoch,
             global_steps,
         )
+    if args.timing_breakdown and comm.is_main_process():
+        log.info(timer.format_summary())
     if train_start is not None and global_steps > start_steps + 1:
         elapsed = time.perf_counter() - train_start
         seq_per_sec = (
