@@ -75,6 +75,30 @@ __device__ __forceinline__ float wave_reduce_max(float v) {
   return v;
 }
 
+// cross-wave sum of an (a,b) pair through LDS; result broadcast to all
+// threads. smem must hold 2 * (blockDim.x/64) floats. Single-use per
+// __shared__ buffer (no trailing barrier).
+__device__ __forceinline__ void block_reduce_sum2(float& a, float& b,
+                                                  float* smem) {
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int nwaves = blockDim.x / WAVE_SIZE;
+  a = wave_reduce_sum(a);
+  b = wave_reduce_sum(b);
+  if (nwaves == 1) return;
+  if (lane == 0) {
+    smem[2 * wave] = a;
+    smem[2 * wave + 1] = b;
+  }
+  __syncthreads();
+  a = 0.f;
+  b = 0.f;
+  for (int w = 0; w < nwaves; ++w) {
+    a += smem[2 * w];
+    b += smem[2 * w + 1];
+  }
+}
+
 // reduce across a block of NW waves through LDS; result broadcast to all.
 template <int NW>
 __device__ __forceinline__ float block_reduce_sum(float v, float* smem) {

@@ -11,11 +11,17 @@
 // padding mask enters as per-sequence valid lengths; dropout by
 // Philox4x32-10 regenerated (not stored) in backward.
 //
-// Backward (flash-2 style): grid (S/64 kv-tiles, B*heads); each block
-// owns one K/V tile, loops over q-tiles; recomputes P from the saved
-// logsumexp; accumulates dK/dV in registers (written once, exclusively,
-// straight into dqkv) and dQ via fp32 atomics into a workspace packed
-// by a final kernel.
+// Backward (flash-2 style, atomic-free): two MFMA kernels.
+// * attn_bwd_kernel: grid (S/64 kv-tiles, B*heads); each block owns one
+//   K/V tile, loops over q-tiles; recomputes P from the saved
+//   logsumexp; accumulates dK/dV in registers, written once,
+//   exclusively, straight into dqkv.
+// * attn_dq_kernel: grid (S/64 q-tiles, B*heads); mirrors the forward's
+//   wave-owns-q structure, recomputes S^T/dP^T per K/V tile and chains
+//   dS^T fragments into dQ += dS.K MFMAs against K^T staged in LDS —
+//   dQ lands in registers and is written once (the previous design
+//   pushed ~1k fp32 atomicAdds per wave per q-iter into an fp32
+//   workspace plus a pack kernel; that was the backward bottleneck).
 //
 // Fragment convention (consistent A/B slot mapping):
 //   A[i][k]/B[k][j]: i|j = lane&15, k = (lane>>4)*4 + (e&3) + 16*(e>>2)
@@ -272,13 +278,160 @@ __global__ void attn_delta_kernel(const __bf16* __restrict__ dout,
   delta[row] = wave_reduce_sum(d);
 }
 
+// dQ kernel: wave-owns-q structure copied from attn_fwd_kernel.
+// Per K/V tile: recompute S^T = K.Q^T and dP^T = V.dO^T via MFMA,
+// P = exp(S*scale - lse), dS^T = P*(dP - delta)*scale, then chain dS^T
+// lane registers into A-fragments for dQ += dS.K against K^T in LDS
+// (exactly how the forward chains P into the PV product). dQ stays in
+// registers until the single epilogue store into dqkv's Q slots.
+template <bool TRAIN_DROP>
+__global__ __launch_bounds__(256) void attn_dq_kernel(
+    const __bf16* __restrict__ dout, const __bf16* __restrict__ qkv,
+    const int* __restrict__ seqlens, const float* __restrict__ lse,
+    const float* __restrict__ delta, __bf16* __restrict__ dqkv, int B, int S,
+    int NH, float p, float scale, uint64_t seed, uint64_t offset) {
+  const int bh = blockIdx.y;
+  const int b = bh / NH, h = bh % NH;
+  const int q0 = blockIdx.x * 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int g = (lane >> 4), li = lane & 15;
+  const int H = NH * 64;
+  const int rs3 = 3 * H;
+  const __bf16* qbase = qkv + static_cast<int64_t>(b) * S * rs3 + h * 64;
+  const __bf16* kbase = qbase + H;
+  const __bf16* vbase = qbase + 2 * H;
+  const __bf16* dobase = dout + static_cast<int64_t>(b) * S * H + h * 64;
+  const int slen = seqlens[b];
+  const float inv_keep = TRAIN_DROP ? 1.f / (1.f - p) : 1.f;
+  Philox philox(seed);
+  const int s2 = S >> 1;
+  const uint64_t drop_base = offset + static_cast<uint64_t>(bh) * s2 * s2;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* K_lds = reinterpret_cast<__bf16*>(smem);   // [64][72] natural
+  __bf16* V_lds = K_lds + 64 * kStride;              // [64][72] natural
+  __bf16* Kt_lds = V_lds + 64 * kStride;             // [64][72] transposed
+
+  // this wave's 16 q rows: Q and dO fragments + lse/delta, registers
+  const int q_row = q0 + wave * 16 + li;
+  const int q_ld = min(q_row, S - 1);
+  bf16x8 qfrag[2], dofrag[2];
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    qfrag[c] = frag_row(qbase + static_cast<int64_t>(q_ld) * rs3, 32 * c, g);
+    dofrag[c] = frag_row(dobase + static_cast<int64_t>(q_ld) * H, 32 * c, g);
+  }
+  const float lse_q = (q_row < S) ? lse[static_cast<int64_t>(bh) * S + q_row] : 0.f;
+  const float dlt_q = (q_row < S) ? delta[static_cast<int64_t>(bh) * S + q_row] : 0.f;
+
+  f32x4 acc_dq[4] = {};
+
+  const int n_kv = (S + 63) / 64;
+  for (int kt = 0; kt < n_kv; ++kt) {
+    const int k0 = kt * 64;
+    __syncthreads();
+    {
+      const int row = tid >> 2, colc = (tid & 3) * 16;
+      const int krow = k0 + row;
+      __bf16 kv[16];
+      if (krow < S) {
+        const uint4* ks = reinterpret_cast<const uint4*>(
+            kbase + static_cast<int64_t>(krow) * rs3 + colc);
+        *reinterpret_cast<uint4*>(&kv[0]) = ks[0];
+        *reinterpret_cast<uint4*>(&kv[8]) = ks[1];
+        const uint4* vs = reinterpret_cast<const uint4*>(
+            vbase + static_cast<int64_t>(krow) * rs3 + colc);
+        *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc]) = vs[0];
+        *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc + 8]) = vs[1];
+      } else {
+        uint4 zero{0, 0, 0, 0};
+#pragma unroll
+        for (int j = 0; j < 16; ++j) kv[j] = __bf16(0.f);
+        *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc]) = zero;
+        *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc + 8]) = zero;
+      }
+      *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc]) =
+          *reinterpret_cast<uint4*>(&kv[0]);
+      *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc + 8]) =
+          *reinterpret_cast<uint4*>(&kv[8]);
+#pragma unroll
+      for (int j = 0; j < 16; ++j) Kt_lds[(colc + j) * kStride + row] = kv[j];
+    }
+    __syncthreads();
+
+    // S^T and dP^T tiles (C[key][q], q = li): A = K rows / V rows
+    float dsv[4][4];
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      const __bf16* krow = &K_lds[(t * 16 + li) * kStride];
+      const __bf16* vrow = &V_lds[(t * 16 + li) * kStride];
+      f32x4 sacc = {}, dpacc = {};
+      sacc = MFMA16(frag_row(krow, 0, g), qfrag[0], sacc);
+      sacc = MFMA16(frag_row(krow, 32, g), qfrag[1], sacc);
+      dpacc = MFMA16(frag_row(vrow, 0, g), dofrag[0], dpacc);
+      dpacc = MFMA16(frag_row(vrow, 32, g), dofrag[1], dpacc);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int key = k0 + t * 16 + g * 4 + r;
+        const bool valid = key < slen && key < S && q_row < S;
+        const float pr =
+            valid ? __expf(sacc[r] * scale - lse_q) : 0.f;
+        float dpd = dpacc[r];
+        if (TRAIN_DROP) {
+          uint32_t r4[4];
+          philox(drop_base + static_cast<uint64_t>(q_row >> 1) * s2 +
+                     (key >> 1),
+                 r4);
+          const bool keep =
+              u32_to_uniform(r4[(q_row & 1) * 2 + (key & 1)]) >= p;
+          dpd = keep ? dpd * inv_keep : 0.f;
+        }
+        dsv[t][r] = pr * (dpd - dlt_q) * scale;
+      }
+    }
+
+    // chain dS^T into A-fragments over keys (chunk c: keys 32c..32c+31)
+    bf16x8 dsfrag[2];
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      union {
+        bf16x8 v;
+        __bf16 e[8];
+      } a;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        a.e[e] = __bf16(dsv[2 * c + (e >> 2)][e & 3]);
+      dsfrag[c] = a.v;
+    }
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      const __bf16* kt_row = &Kt_lds[(n * 16 + li) * kStride];
+      acc_dq[n] = MFMA16(dsfrag[0], frag_row(kt_row, 0, g), acc_dq[n]);
+      acc_dq[n] = MFMA16(dsfrag[1], frag_row(kt_row, 32, g), acc_dq[n]);
+    }
+  }
+
+  // epilogue: one store per element into dqkv Q slots
+#pragma unroll
+  for (int n = 0; n < 4; ++n) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qr = q0 + wave * 16 + g * 4 + r;
+      if (qr < S) {
+        dqkv[(static_cast<int64_t>(b) * S + qr) * rs3 + h * 64 + n * 16 + li] =
+            __bf16(acc_dq[n][r]);
+      }
+    }
+  }
+}
+
 template <bool TRAIN_DROP>
 __global__ __launch_bounds__(256) void attn_bwd_kernel(
     const __bf16* __restrict__ dout, const __bf16* __restrict__ qkv,
     const int* __restrict__ seqlens, const float* __restrict__ lse,
-    const float* __restrict__ delta, __bf16* __restrict__ dqkv,
-    float* __restrict__ dq_ws, int B, int S, int NH, float p, float scale,
-    uint64_t seed, uint64_t offset) {
+    const float* __restrict__ delta, __bf16* __restrict__ dqkv, int B, int S,
+    int NH, float p, float scale, uint64_t seed, uint64_t offset) {
   const int bh = blockIdx.y;
   const int b = bh / NH, h = bh % NH;
   const int k0 = blockIdx.x * 64;
@@ -299,17 +452,15 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* K_lds = reinterpret_cast<__bf16*>(smem);   // [64][72] natural
-  __bf16* Kt_lds = K_lds + 64 * kStride;             // [64][72] transposed
-  __bf16* V_lds = Kt_lds + 64 * kStride;             // [64][72] natural
+  __bf16* V_lds = K_lds + 64 * kStride;              // [64][72] natural
   __bf16* Q_lds = V_lds + 64 * kStride;              // [64][72] natural
   __bf16* Qt_lds = Q_lds + 64 * kStride;             // [64][72] transposed
   __bf16* dO_lds = Qt_lds + 64 * kStride;            // [64][72] natural
   __bf16* dOt_lds = dO_lds + 64 * kStride;           // [64][72] transposed
-  __bf16* dS_lds = dOt_lds + 64 * kStride;           // [64][72]
-  float* lse_lds = reinterpret_cast<float*>(dS_lds + 64 * kStride);  // [64]
-  float* dlt_lds = lse_lds + 64;                                     // [64]
+  float* lse_lds = reinterpret_cast<float*>(dOt_lds + 64 * kStride);  // [64]
+  float* dlt_lds = lse_lds + 64;                                      // [64]
 
-  // stage K (natural + transposed) and V (natural) once
+  // stage K and V (natural) once
   {
     const int row = tid >> 2, colc = (tid & 3) * 16;
     const int krow = k0 + row;
@@ -335,8 +486,6 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
         *reinterpret_cast<uint4*>(&vv[0]);
     *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc + 8]) =
         *reinterpret_cast<uint4*>(&vv[8]);
-#pragma unroll
-    for (int j = 0; j < 16; ++j) Kt_lds[(colc + j) * kStride + row] = kv[j];
   }
   __syncthreads();
 
@@ -431,14 +580,6 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       }
     }
 
-    // dS -> LDS (for the dQ pass), bf16
-#pragma unroll
-    for (int mq = 0; mq < 4; ++mq)
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        dS_lds[(mq * 16 + g * 4 + r) * kStride + key_local] =
-            __bf16(dsv[mq][r]);
-
     // chain P and dS registers into B-fragments over q (chunk c: q 32c+..)
     bf16x8 pfrag[2], dsfrag[2];
 #pragma unroll
@@ -466,29 +607,6 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       dk_acc[m] = MFMA16(frag_row(qt_row, 0, g), dsfrag[0], dk_acc[m]);
       dk_acc[m] = MFMA16(frag_row(qt_row, 32, g), dsfrag[1], dk_acc[m]);
     }
-    __syncthreads();  // dS_lds complete across waves
-
-    // dQ: wave owns q-subtile `wave`; A = dS rows, B = K^T (Kt rows)
-    {
-      const __bf16* ds_row = &dS_lds[(wave * 16 + li) * kStride];
-      bf16x8 a0 = frag_row(ds_row, 0, g);
-      bf16x8 a1 = frag_row(ds_row, 32, g);
-#pragma unroll
-      for (int n = 0; n < 4; ++n) {
-        const __bf16* kt_row = &Kt_lds[(n * 16 + li) * kStride];
-        f32x4 dq = {};
-        dq = MFMA16(a0, frag_row(kt_row, 0, g), dq);
-        dq = MFMA16(a1, frag_row(kt_row, 32, g), dq);
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int q_abs = q0 + wave * 16 + g * 4 + r;
-          if (q_abs < S)
-            atomicAdd(&dq_ws[(static_cast<int64_t>(bh) * S + q_abs) * 64 +
-                             n * 16 + li],
-                      dq[r]);
-        }
-      }
-    }
   }
 
   // write dK/dV straight into dqkv (this block owns keys k0..k0+63)
@@ -504,24 +622,6 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
         dqkv[rowb * rs3 + 2 * H + h * 64 + dh] = __bf16(dv_acc[m][r]);
       }
     }
-  }
-}
-
-// pack fp32 dQ workspace into the bf16 dqkv Q slots
-__global__ void attn_pack_dq_kernel(const float* __restrict__ dq_ws,
-                                    __bf16* __restrict__ dqkv, int B, int S,
-                                    int NH) {
-  const int H = NH * 64;
-  const int64_t total = static_cast<int64_t>(B) * NH * S * 64;
-  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += static_cast<int64_t>(gridDim.x) * blockDim.x) {
-    const int d = static_cast<int>(i & 63);
-    const int64_t rest = i >> 6;
-    const int q = static_cast<int>(rest % S);
-    const int64_t bh = rest / S;
-    const int b = static_cast<int>(bh) / NH, h = static_cast<int>(bh) % NH;
-    dqkv[(static_cast<int64_t>(b) * S + q) * 3 * H + h * 64 + d] =
-        __bf16(dq_ws[i]);
   }
 }
 
@@ -578,7 +678,6 @@ torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
   auto seql = seqlens.to(qkv.device(), torch::kInt32).contiguous();
   auto dqkv = torch::empty_like(qkv);
   auto fopts = qkv.options().dtype(torch::kFloat32);
-  auto dq_ws = torch::zeros({static_cast<int64_t>(B) * NH * S * 64}, fopts);
   auto delta = torch::empty({static_cast<int64_t>(B) * NH * S}, fopts);
   auto stream = at::hip::getCurrentHIPStream();
   auto dout_c = dout.contiguous();
@@ -591,44 +690,28 @@ torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
                      delta.data_ptr<float>(), B, S, NH);
 
   dim3 grid((S + 63) / 64, B * NH), block(256);
-  const size_t lds = 8 * 64 * kStride * sizeof(__bf16) + 2 * 64 * sizeof(float);
-  // backward wants ~74 KB LDS (> the 64 KB default cap; MI355X has 160)
-  HIP_CHECK(hipFuncSetAttribute(
-      reinterpret_cast<const void*>(&attn_bwd_kernel<true>),
-      hipFuncAttributeMaxDynamicSharedMemorySize, lds));
-  HIP_CHECK(hipFuncSetAttribute(
-      reinterpret_cast<const void*>(&attn_bwd_kernel<false>),
-      hipFuncAttributeMaxDynamicSharedMemorySize, lds));
+  const size_t lds = 6 * 64 * kStride * sizeof(__bf16) + 2 * 64 * sizeof(float);
+  const size_t lds_dq = 3 * 64 * kStride * sizeof(__bf16);
   const float scale = 1.0f / sqrtf(64.f);
   const bool train_drop = p > 0.0;
+  auto args = [&](auto kernel, size_t lds_bytes) {
+    hipLaunchKernelGGL(kernel, grid, block, lds_bytes, stream,
+                       reinterpret_cast<const __bf16*>(dout_c.data_ptr()),
+                       reinterpret_cast<const __bf16*>(qkv.data_ptr()),
+                       seql.data_ptr<int>(), lse.data_ptr<float>(),
+                       delta.data_ptr<float>(),
+                       reinterpret_cast<__bf16*>(dqkv.data_ptr()), B, S, NH,
+                       static_cast<float>(p), scale,
+                       static_cast<uint64_t>(seed),
+                       static_cast<uint64_t>(offset));
+  };
   if (train_drop) {
-    hipLaunchKernelGGL((attn_bwd_kernel<true>), grid, block, lds, stream,
-                       reinterpret_cast<const __bf16*>(dout_c.data_ptr()),
-                       reinterpret_cast<const __bf16*>(qkv.data_ptr()),
-                       seql.data_ptr<int>(), lse.data_ptr<float>(),
-                       delta.data_ptr<float>(),
-                       reinterpret_cast<__bf16*>(dqkv.data_ptr()),
-                       dq_ws.data_ptr<float>(), B, S, NH,
-                       static_cast<float>(p), scale,
-                       static_cast<uint64_t>(seed),
-                       static_cast<uint64_t>(offset));
+    args(attn_bwd_kernel<true>, lds);
+    args(attn_dq_kernel<true>, lds_dq);
   } else {
-    hipLaunchKernelGGL((attn_bwd_kernel<false>), grid, block, lds, stream,
-                       reinterpret_cast<const __bf16*>(dout_c.data_ptr()),
-                       reinterpret_cast<const __bf16*>(qkv.data_ptr()),
-                       seql.data_ptr<int>(), lse.data_ptr<float>(),
-                       delta.data_ptr<float>(),
-                       reinterpret_cast<__bf16*>(dqkv.data_ptr()),
-                       dq_ws.data_ptr<float>(), B, S, NH,
-                       static_cast<float>(p), scale,
-                       static_cast<uint64_t>(seed),
-                       static_cast<uint64_t>(offset));
+    args(attn_bwd_kernel<false>, lds);
+    args(attn_dq_kernel<false>, lds_dq);
   }
-  const int64_t total = rows * 64;
-  const int pblocks = static_cast<int>(std::min<int64_t>((total + 255) / 256, 2048));
-  hipLaunchKernelGGL(attn_pack_dq_kernel, dim3(pblocks), dim3(256), 0, stream,
-                     dq_ws.data_ptr<float>(),
-                     reinterpret_cast<__bf16*>(dqkv.data_ptr()), B, S, NH);
   return dqkv;
 }
 

@@ -3,8 +3,17 @@
 // One kernel for the residual junction the reference runs as 4 ops
 // (BertSelfOutput/BertOutput: src/modeling.py:432-443, 468-479):
 //   z = dropout(x + bias) + residual ; y = LN(z) * gamma + beta
-// Dropout uses Philox4x32-10 (seed, offset) and stores the keep-mask as
-// bytes so backward is exact. One wave per row, 16 B/lane vector IO.
+// Dropout uses Philox4x32-10 (seed, offset) keyed by element index, and
+// stores the keep-mask as bytes so backward is exact.
+//
+// Memory-bound design (same as layernorm.hip):
+// * forward: one BLOCK per row; x/residual loaded once into registers,
+//   z kept in registers through the mean/var block reduction, y written
+//   from registers — no re-read of z after the reduction.
+// * backward: one WAVE per row stripe, dy/z register-resident across
+//   both passes, dgamma/dbeta/dbias accumulated in per-lane VGPRs
+//   across the stripe (no per-element LDS read-modify-write), one
+//   [n_waves, 3H] partial buffer reduced by a single col_reduce launch.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -16,7 +25,7 @@ namespace bpa {
 __global__ void col_reduce_kernel(const float* __restrict__ parts, int nparts,
                                   int H, float* __restrict__ out);
 
-template <typename T, int VEC, bool HAS_BIAS, bool TRAIN_DROP>
+template <typename T, int VEC, int ITEMS, bool HAS_BIAS, bool TRAIN_DROP>
 __global__ void bdrl_fwd_kernel(
     const T* __restrict__ x, const float* __restrict__ bias,
     const T* __restrict__ residual, const float* __restrict__ gamma,
@@ -24,139 +33,143 @@ __global__ void bdrl_fwd_kernel(
     uint8_t* __restrict__ mask, float* __restrict__ mean,
     float* __restrict__ rstd, int rows, int H, float p, float eps,
     uint64_t seed, uint64_t offset) {
-  const int lane = threadIdx.x & (WAVE_SIZE - 1);
-  const int wave = threadIdx.x / WAVE_SIZE;
-  const int row = blockIdx.x * (blockDim.x / WAVE_SIZE) + wave;
+  __shared__ float red[32];
+  const int row = blockIdx.x;
   if (row >= rows) return;
   const int64_t base = static_cast<int64_t>(row) * H;
   const float keep_scale = TRAIN_DROP ? 1.0f / (1.0f - p) : 1.0f;
   Philox philox(seed);
 
+  float zv[ITEMS][VEC];
+  int cols[ITEMS];
   float sum = 0.f, sumsq = 0.f;
-  for (int c = lane * VEC; c < H; c += WAVE_SIZE * VEC) {
-    T xv[VEC], rv[VEC], zv[VEC];
-    *reinterpret_cast<uint4*>(xv) = *reinterpret_cast<const uint4*>(x + base + c);
-    *reinterpret_cast<uint4*>(rv) =
-        *reinterpret_cast<const uint4*>(residual + base + c);
-    uint8_t mv[VEC];
-    if (TRAIN_DROP) {
-      // 4 uniforms per philox call; VEC=8 -> 2 calls, VEC=4 -> 1 call
 #pragma unroll
-      for (int q = 0; q < VEC / 4; ++q) {
-        uint32_t r4[4];
-        philox(offset + (base + c) / 4 + q, r4);
+  for (int i = 0; i < ITEMS; ++i) {
+    const int c = (i * blockDim.x + threadIdx.x) * VEC;
+    cols[i] = c;
+    if (c < H) {
+      T xv[VEC], rv[VEC], zt[VEC];
+      *reinterpret_cast<uint4*>(xv) = *reinterpret_cast<const uint4*>(x + base + c);
+      *reinterpret_cast<uint4*>(rv) =
+          *reinterpret_cast<const uint4*>(residual + base + c);
+      uint8_t mv[VEC];
+      if (TRAIN_DROP) {
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
-          mv[q * 4 + j] = u32_to_uniform(r4[j]) >= p ? 1 : 0;
+        for (int q = 0; q < VEC / 4; ++q) {
+          uint32_t r4[4];
+          philox(offset + (base + c) / 4 + q, r4);
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            mv[q * 4 + j] = u32_to_uniform(r4[j]) >= p ? 1 : 0;
+        }
+      }
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        float t = DTraits<T>::to_f32(xv[k]);
+        if (HAS_BIAS) t += bias[c + k];
+        if (TRAIN_DROP) t = mv[k] ? t * keep_scale : 0.f;
+        t += DTraits<T>::to_f32(rv[k]);
+        zv[i][k] = t;
+        zt[k] = DTraits<T>::from_f32(t);
+        sum += t;
+        sumsq += t * t;
+      }
+      *reinterpret_cast<uint4*>(z + base + c) = *reinterpret_cast<const uint4*>(zt);
+      if (TRAIN_DROP) {
+        if (VEC == 8)
+          *reinterpret_cast<uint2*>(mask + base + c) =
+              *reinterpret_cast<const uint2*>(mv);
+        else
+          *reinterpret_cast<uint32_t*>(mask + base + c) =
+              *reinterpret_cast<const uint32_t*>(mv);
       }
     }
-#pragma unroll
-    for (int k = 0; k < VEC; ++k) {
-      float t = DTraits<T>::to_f32(xv[k]);
-      if (HAS_BIAS) t += bias[c + k];
-      if (TRAIN_DROP) t = mv[k] ? t * keep_scale : 0.f;
-      t += DTraits<T>::to_f32(rv[k]);
-      zv[k] = DTraits<T>::from_f32(t);
-      sum += t;
-      sumsq += t * t;
-    }
-    *reinterpret_cast<uint4*>(z + base + c) = *reinterpret_cast<const uint4*>(zv);
-    if (TRAIN_DROP) {
-      if (VEC == 8)
-        *reinterpret_cast<uint2*>(mask + base + c) =
-            *reinterpret_cast<const uint2*>(mv);
-      else
-        *reinterpret_cast<uint32_t*>(mask + base + c) =
-            *reinterpret_cast<const uint32_t*>(mv);
-    }
   }
-  sum = wave_reduce_sum(sum);
-  sumsq = wave_reduce_sum(sumsq);
+  block_reduce_sum2(sum, sumsq, red);
   const float mu = sum / H;
   const float var = fmaxf(sumsq / H - mu * mu, 0.f);
   const float rs = rsqrtf(var + eps);
-  if (lane == 0) {
+  if (threadIdx.x == 0) {
     mean[row] = mu;
     rstd[row] = rs;
   }
-  for (int c = lane * VEC; c < H; c += WAVE_SIZE * VEC) {
-    T zv[VEC], ov[VEC];
-    *reinterpret_cast<uint4*>(zv) = *reinterpret_cast<const uint4*>(z + base + c);
 #pragma unroll
-    for (int k = 0; k < VEC; ++k) {
-      float t = DTraits<T>::to_f32(zv[k]);
-      ov[k] = DTraits<T>::from_f32((t - mu) * rs * gamma[c + k] + beta[c + k]);
+  for (int i = 0; i < ITEMS; ++i) {
+    const int c = cols[i];
+    if (c < H) {
+      T ov[VEC];
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        ov[k] = DTraits<T>::from_f32((zv[i][k] - mu) * rs * gamma[c + k] +
+                                     beta[c + k]);
+      }
+      *reinterpret_cast<uint4*>(y + base + c) = *reinterpret_cast<const uint4*>(ov);
     }
-    *reinterpret_cast<uint4*>(y + base + c) = *reinterpret_cast<const uint4*>(ov);
   }
 }
 
-template <typename T, int VEC, int NW, bool HAS_BIAS, bool TRAIN_DROP>
-__global__ void bdrl_bwd_kernel(
+// backward dx/dz: one BLOCK per row (mirror of bdrl_fwd_kernel); dy/z/
+// mask loaded once into registers, s1/s2 via block reduce, dz_res and
+// dx written from registers. dgamma/dbeta/dbias partials come from the
+// shared streaming col_stats_kernel (layernorm.hip) reading dy, z and
+// the dx this kernel wrote.
+template <typename T, int VEC, int ITEMS, bool TRAIN_DROP>
+__global__ void bdrl_bwd_dx_kernel(
     const T* __restrict__ dy, const T* __restrict__ z,
     const uint8_t* __restrict__ mask, const float* __restrict__ gamma,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    T* __restrict__ dx, T* __restrict__ dz_res,
-    float* __restrict__ part_dgamma, float* __restrict__ part_dbeta,
-    float* __restrict__ part_dbias, int rows, int H, float p,
-    int rows_per_block) {
-  const int lane = threadIdx.x & (WAVE_SIZE - 1);
-  const int wave = threadIdx.x / WAVE_SIZE;
-  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  const int n_slabs = HAS_BIAS ? 3 : 2;
-  // per-wave private slabs [NW][n_slabs][H]: no atomics, no contention
-  float* lg = reinterpret_cast<float*>(smem_raw) + wave * n_slabs * H;
-  float* lb = lg + H;
-  float* lbias = lb + H;
-  for (int c = lane; c < n_slabs * H; c += WAVE_SIZE) lg[c] = 0.f;
+    T* __restrict__ dx, T* __restrict__ dz_res, int rows, int H, float p) {
+  __shared__ float red[32];
+  const int row = blockIdx.x;
+  if (row >= rows) return;
+  const int64_t base = static_cast<int64_t>(row) * H;
+  const float mu = mean[row], rs = rstd[row];
   const float keep_scale = TRAIN_DROP ? 1.0f / (1.0f - p) : 1.0f;
 
-  const int row0 = blockIdx.x * rows_per_block;
-  const int row_end = min(row0 + rows_per_block, rows);
-  for (int r = row0 + wave; r < row_end; r += NW) {
-    const int64_t base = static_cast<int64_t>(r) * H;
-    const float mu = mean[r], rs = rstd[r];
-    float s1 = 0.f, s2 = 0.f;
-    for (int c = lane * VEC; c < H; c += WAVE_SIZE * VEC) {
-      T dv[VEC], zv[VEC];
-      *reinterpret_cast<uint4*>(dv) = *reinterpret_cast<const uint4*>(dy + base + c);
-      *reinterpret_cast<uint4*>(zv) = *reinterpret_cast<const uint4*>(z + base + c);
+  float dw[ITEMS][VEC], zh[ITEMS][VEC];
+  uint8_t mv[ITEMS][VEC];
+  int cols[ITEMS];
+  float s1 = 0.f, s2 = 0.f;
 #pragma unroll
-      for (int k = 0; k < VEC; ++k) {
-        float d = DTraits<T>::to_f32(dv[k]);
-        float zh = (DTraits<T>::to_f32(zv[k]) - mu) * rs;
-        float dw = d * gamma[c + k];
-        s1 += dw * zh;
-        s2 += dw;
-        lg[c + k] += d * zh;
-        lb[c + k] += d;
-      }
-    }
-    s1 = wave_reduce_sum(s1) / H;
-    s2 = wave_reduce_sum(s2) / H;
-
-    for (int c = lane * VEC; c < H; c += WAVE_SIZE * VEC) {
-      T dv[VEC], zv[VEC], dzo[VEC], dxo[VEC];
-      *reinterpret_cast<uint4*>(dv) = *reinterpret_cast<const uint4*>(dy + base + c);
-      *reinterpret_cast<uint4*>(zv) = *reinterpret_cast<const uint4*>(z + base + c);
-      uint8_t mv[VEC];
+  for (int i = 0; i < ITEMS; ++i) {
+    const int c = (i * blockDim.x + threadIdx.x) * VEC;
+    cols[i] = c;
+    if (c < H) {
+      T dt[VEC], zt[VEC];
+      *reinterpret_cast<uint4*>(dt) = *reinterpret_cast<const uint4*>(dy + base + c);
+      *reinterpret_cast<uint4*>(zt) = *reinterpret_cast<const uint4*>(z + base + c);
       if (TRAIN_DROP) {
         if (VEC == 8)
-          *reinterpret_cast<uint2*>(mv) =
+          *reinterpret_cast<uint2*>(mv[i]) =
               *reinterpret_cast<const uint2*>(mask + base + c);
         else
-          *reinterpret_cast<uint32_t*>(mv) =
+          *reinterpret_cast<uint32_t*>(mv[i]) =
               *reinterpret_cast<const uint32_t*>(mask + base + c);
       }
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
-        float d = DTraits<T>::to_f32(dv[k]);
-        float zh = (DTraits<T>::to_f32(zv[k]) - mu) * rs;
-        float dzk = rs * (d * gamma[c + k] - s2 - zh * s1);
+        zh[i][k] = (DTraits<T>::to_f32(zt[k]) - mu) * rs;
+        dw[i][k] = DTraits<T>::to_f32(dt[k]) * gamma[c + k];
+        s1 += dw[i][k] * zh[i][k];
+        s2 += dw[i][k];
+      }
+    }
+  }
+  block_reduce_sum2(s1, s2, red);
+  s1 /= H;
+  s2 /= H;
+#pragma unroll
+  for (int i = 0; i < ITEMS; ++i) {
+    const int c = cols[i];
+    if (c < H) {
+      T dzo[VEC], dxo[VEC];
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        const float dzk = rs * (dw[i][k] - s2 - zh[i][k] * s1);
         dzo[k] = DTraits<T>::from_f32(dzk);  // grad to residual input
-        float dxk = TRAIN_DROP ? (mv[k] ? dzk * keep_scale : 0.f) : dzk;
+        const float dxk =
+            TRAIN_DROP ? (mv[i][k] ? dzk * keep_scale : 0.f) : dzk;
         dxo[k] = DTraits<T>::from_f32(dxk);
-        if (HAS_BIAS) lbias[c + k] += dxk;
       }
       *reinterpret_cast<uint4*>(dz_res + base + c) =
           *reinterpret_cast<const uint4*>(dzo);
@@ -164,22 +177,14 @@ __global__ void bdrl_bwd_kernel(
           *reinterpret_cast<const uint4*>(dxo);
     }
   }
-  __syncthreads();
-  float* slab0 = reinterpret_cast<float*>(smem_raw);
-  for (int c = threadIdx.x; c < H; c += blockDim.x) {
-    float ag = 0.f, ab = 0.f, abias = 0.f;
-#pragma unroll
-    for (int w = 0; w < NW; ++w) {
-      ag += slab0[w * n_slabs * H + c];
-      ab += slab0[w * n_slabs * H + H + c];
-      if (HAS_BIAS) abias += slab0[w * n_slabs * H + 2 * H + c];
-    }
-    part_dgamma[static_cast<int64_t>(blockIdx.x) * H + c] = ag;
-    part_dbeta[static_cast<int64_t>(blockIdx.x) * H + c] = ab;
-    if (HAS_BIAS)
-      part_dbias[static_cast<int64_t>(blockIdx.x) * H + c] = abias;
-  }
 }
+
+// from layernorm.hip
+template <typename T, int VEC>
+void launch_col_stats(const T* dy, const T* z, const T* dx_src,
+                      const float* mean, const float* rstd, float* part,
+                      int rows, int H, int rows_per_chunk, int n_chunks,
+                      bool has_bias, hipStream_t stream);
 
 #define DISPATCH_T(TYPE, NAME, ...)                                          \
   [&] {                                                                      \
@@ -225,15 +230,16 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
   auto mean = torch::empty({rows}, fopts);
   auto rstd = torch::empty({rows}, fopts);
   auto stream = at::hip::getCurrentHIPStream();
-  constexpr int WAVES = 4;
-  dim3 grid((rows + WAVES - 1) / WAVES), block(WAVES * WAVE_SIZE);
   DISPATCH_T(x.scalar_type(), "bdrl_fwd", [&] {
     TORCH_CHECK(H % kVec == 0, "bdrl_fwd: H % ", kVec, " != 0");
-    auto launch = [&](auto has_bias_c, auto train_c) {
+    const int slices = H / kVec;
+    auto launch = [&](auto items_c, int threads, auto has_bias_c,
+                      auto train_c) {
       hipLaunchKernelGGL(
-          (bdrl_fwd_kernel<scalar_t, kVec, decltype(has_bias_c)::value,
+          (bdrl_fwd_kernel<scalar_t, kVec, decltype(items_c)::value,
+                           decltype(has_bias_c)::value,
                            decltype(train_c)::value>),
-          grid, block, 0, stream,
+          dim3(rows), dim3(threads), 0, stream,
           reinterpret_cast<const scalar_t*>(x.data_ptr()),
           has_bias ? bias_f.data_ptr<float>() : nullptr,
           reinterpret_cast<const scalar_t*>(res_c.data_ptr()),
@@ -245,14 +251,28 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
           static_cast<float>(p), static_cast<float>(eps),
           static_cast<uint64_t>(seed), static_cast<uint64_t>(offset));
     };
-    if (has_bias && train_drop)
-      launch(std::true_type{}, std::true_type{});
-    else if (has_bias)
-      launch(std::true_type{}, std::false_type{});
-    else if (train_drop)
-      launch(std::false_type{}, std::true_type{});
-    else
-      launch(std::false_type{}, std::false_type{});
+    auto by_flags = [&](auto items_c, int threads) {
+      if (has_bias && train_drop)
+        launch(items_c, threads, std::true_type{}, std::true_type{});
+      else if (has_bias)
+        launch(items_c, threads, std::true_type{}, std::false_type{});
+      else if (train_drop)
+        launch(items_c, threads, std::false_type{}, std::true_type{});
+      else
+        launch(items_c, threads, std::false_type{}, std::false_type{});
+    };
+    if (slices <= 256)
+      by_flags(std::integral_constant<int, 1>{}, ((slices + 63) / 64) * 64);
+    else if (slices <= 512)
+      by_flags(std::integral_constant<int, 2>{}, 256);
+    else if (slices <= 1024)
+      by_flags(std::integral_constant<int, 2>{}, 512);
+    else if (slices <= 2048)
+      by_flags(std::integral_constant<int, 4>{}, 512);
+    else {
+      TORCH_CHECK(slices <= 4096, "bdrl_fwd: H too large");
+      by_flags(std::integral_constant<int, 4>{}, 1024);
+    }
   });
   return {y, z, mask, mean, rstd};
 }
@@ -265,69 +285,67 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
   auto dy_c = dy.contiguous();
   auto dx = torch::empty_like(z);
   auto dz_res = torch::empty_like(z);
-  constexpr int NW = 4;
-  const int rows_per_block = 16;
-  const int nblocks = (rows + rows_per_block - 1) / rows_per_block;
+  const int rows_per_chunk = 8;
+  const int n_chunks = (rows + rows_per_chunk - 1) / rows_per_chunk;
+  const int n_slabs = has_bias ? 3 : 2;
   auto fopts = z.options().dtype(torch::kFloat32);
-  auto part_g = torch::empty({nblocks, H}, fopts);
-  auto part_b = torch::empty({nblocks, H}, fopts);
-  auto part_bias =
-      has_bias ? torch::empty({nblocks, H}, fopts) : torch::empty({0}, fopts);
+  auto part = torch::empty({n_chunks, n_slabs * H}, fopts);
+  auto out = torch::zeros({n_slabs * H}, fopts);
   const bool train_drop = p > 0.0;
   auto stream = at::hip::getCurrentHIPStream();
-  const size_t lds =
-      NW * (has_bias ? 3 : 2) * static_cast<size_t>(H) * sizeof(float);
-  TORCH_CHECK(lds <= 160 * 1024, "bdrl_bwd: H too large");
   DISPATCH_T(z.scalar_type(), "bdrl_bwd", [&] {
     TORCH_CHECK(H % kVec == 0, "bdrl_bwd: H % ", kVec, " != 0");
-    auto launch = [&](auto has_bias_c, auto train_c) {
-      if (lds > 48 * 1024) {
-        HIP_CHECK(hipFuncSetAttribute(
-            reinterpret_cast<const void*>(
-                &bdrl_bwd_kernel<scalar_t, kVec, NW,
-                                 decltype(has_bias_c)::value,
-                                 decltype(train_c)::value>),
-            hipFuncAttributeMaxDynamicSharedMemorySize, lds));
-      }
+    const int slices = H / kVec;
+    auto launch = [&](auto items_c, int threads, auto train_c) {
       hipLaunchKernelGGL(
-          (bdrl_bwd_kernel<scalar_t, kVec, NW, decltype(has_bias_c)::value,
-                           decltype(train_c)::value>),
-          dim3(nblocks), dim3(NW * WAVE_SIZE), lds, stream,
+          (bdrl_bwd_dx_kernel<scalar_t, kVec, decltype(items_c)::value,
+                              decltype(train_c)::value>),
+          dim3(rows), dim3(threads), 0, stream,
           reinterpret_cast<const scalar_t*>(dy_c.data_ptr()),
           reinterpret_cast<const scalar_t*>(z.data_ptr()),
           train_drop ? mask.data_ptr<uint8_t>() : nullptr,
           gamma_f.data_ptr<float>(), mean.data_ptr<float>(),
           rstd.data_ptr<float>(),
           reinterpret_cast<scalar_t*>(dx.data_ptr()),
-          reinterpret_cast<scalar_t*>(dz_res.data_ptr()),
-          part_g.data_ptr<float>(), part_b.data_ptr<float>(),
-          has_bias ? part_bias.data_ptr<float>() : nullptr, rows, H,
-          static_cast<float>(p), rows_per_block);
+          reinterpret_cast<scalar_t*>(dz_res.data_ptr()), rows, H,
+          static_cast<float>(p));
     };
-    if (has_bias && train_drop)
-      launch(std::true_type{}, std::true_type{});
-    else if (has_bias)
-      launch(std::true_type{}, std::false_type{});
-    else if (train_drop)
-      launch(std::false_type{}, std::true_type{});
+    auto by_items = [&](auto train_c) {
+      if (slices <= 256)
+        launch(std::integral_constant<int, 1>{},
+               ((slices + 63) / 64) * 64, train_c);
+      else if (slices <= 512)
+        launch(std::integral_constant<int, 2>{}, 256, train_c);
+      else if (slices <= 1024)
+        launch(std::integral_constant<int, 2>{}, 512, train_c);
+      else if (slices <= 2048)
+        launch(std::integral_constant<int, 4>{}, 512, train_c);
+      else {
+        TORCH_CHECK(slices <= 4096, "bdrl_bwd: H too large");
+        launch(std::integral_constant<int, 4>{}, 1024, train_c);
+      }
+    };
+    if (train_drop)
+      by_items(std::true_type{});
     else
-      launch(std::false_type{}, std::false_type{});
+      by_items(std::false_type{});
+    launch_col_stats<scalar_t, kVec>(
+        reinterpret_cast<const scalar_t*>(dy_c.data_ptr()),
+        reinterpret_cast<const scalar_t*>(z.data_ptr()),
+        reinterpret_cast<const scalar_t*>(dx.data_ptr()),
+        mean.data_ptr<float>(), rstd.data_ptr<float>(),
+        part.data_ptr<float>(), rows, H, rows_per_chunk, n_chunks, has_bias,
+        stream);
   });
-  auto dgamma = torch::zeros({H}, fopts);
-  auto dbeta = torch::zeros({H}, fopts);
-  auto dbias = has_bias ? torch::zeros({H}, fopts) : torch::empty({0}, fopts);
-  dim3 rgrid((H + 255) / 256, (nblocks + kColChunk - 1) / kColChunk), rblock(256);
-  hipLaunchKernelGGL(col_reduce_kernel, rgrid, rblock, 0, stream,
-                     part_g.data_ptr<float>(), nblocks, H,
-                     dgamma.data_ptr<float>());
-  hipLaunchKernelGGL(col_reduce_kernel, rgrid, rblock, 0, stream,
-                     part_b.data_ptr<float>(), nblocks, H,
-                     dbeta.data_ptr<float>());
-  if (has_bias) {
-    hipLaunchKernelGGL(col_reduce_kernel, rgrid, rblock, 0, stream,
-                       part_bias.data_ptr<float>(), nblocks, H,
-                       dbias.data_ptr<float>());
-  }
+  dim3 rgrid((n_slabs * H + 255) / 256,
+             (n_chunks + kColChunk - 1) / kColChunk);
+  hipLaunchKernelGGL(col_reduce_kernel, rgrid, dim3(256), 0, stream,
+                     part.data_ptr<float>(), n_chunks, n_slabs * H,
+                     out.data_ptr<float>());
+  auto dgamma = out.narrow(0, 0, H).contiguous();
+  auto dbeta = out.narrow(0, H, H).contiguous();
+  auto dbias =
+      has_bias ? out.narrow(0, 2 * H, H).contiguous() : torch::empty({0}, fopts);
   if (gamma.scalar_type() != torch::kFloat32) {
     dgamma = dgamma.to(gamma.scalar_type());
     dbeta = dbeta.to(gamma.scalar_type());

@@ -1,11 +1,20 @@
 // Fused LayerNorm forward/backward for MI355X (gfx950).
 //
 // Replaces Apex FusedLayerNormAffineFunction (reference call sites:
-// src/modeling.py:299-335). Shapes in this framework: rows up to ~12k
-// (B*S), H in {64..4096}, eps 1e-12. Memory-bound: one wave per row,
-// bf16x8 (16 B/lane) vector loads, mean/var by 64-lane shuffle
-// reduction; backward dgamma/dbeta use a deterministic two-stage
-// partial-sum reduction (no atomics).
+// src/modeling.py:299-335). Shapes here: rows up to ~12k (B*S), H in
+// {64..4096}, eps 1e-12. Memory-bound design:
+//
+// * forward: one BLOCK per row; every lane loads its 16 B slice ONCE
+//   into registers, mean/var via wave shuffle + tiny LDS cross-wave
+//   combine, normalized output written from registers — exactly one
+//   read + one write of x/y (the old one-wave-per-row version re-read
+//   x after the reduction and measured 0.89 TB/s; this shape is pure
+//   streaming).
+// * backward: one WAVE per row stripe with column-fixed lanes; dy/x are
+//   held in registers across the two per-row passes, and dgamma/dbeta
+//   accumulate in per-lane VGPRs across the stripe's rows (the old
+//   version did a per-element LDS read-modify-write). Waves write
+//   [n_waves, 2H] partials, reduced by ONE col_reduce_kernel launch.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -14,129 +23,189 @@
 
 namespace bpa {
 
-template <typename T, int VEC>
+// ITEMS = uint4 slices each thread owns (ITEMS*VEC*blockDim.x >= H).
+template <typename T, int VEC, int ITEMS>
 __global__ void ln_fwd_kernel(const T* __restrict__ x,
                               const float* __restrict__ gamma,
                               const float* __restrict__ beta,
                               T* __restrict__ y, float* __restrict__ mean,
                               float* __restrict__ rstd, int rows, int H,
                               float eps) {
-  const int lane = threadIdx.x & (WAVE_SIZE - 1);
-  const int wave = threadIdx.x / WAVE_SIZE;
-  const int row = blockIdx.x * (blockDim.x / WAVE_SIZE) + wave;
+  __shared__ float red[32];
+  const int row = blockIdx.x;
   if (row >= rows) return;
-  const T* xr = x + static_cast<int64_t>(row) * H;
-  T* yr = y + static_cast<int64_t>(row) * H;
+  const int64_t base = static_cast<int64_t>(row) * H;
 
+  float v[ITEMS][VEC];
+  int cols[ITEMS];
   float sum = 0.f, sumsq = 0.f;
-  for (int c = lane * VEC; c < H; c += WAVE_SIZE * VEC) {
-    T v[VEC];
-    *reinterpret_cast<uint4*>(v) = *reinterpret_cast<const uint4*>(xr + c);
 #pragma unroll
-    for (int k = 0; k < VEC; ++k) {
-      float f = DTraits<T>::to_f32(v[k]);
-      sum += f;
-      sumsq += f * f;
+  for (int i = 0; i < ITEMS; ++i) {
+    const int c = (i * blockDim.x + threadIdx.x) * VEC;
+    cols[i] = c;
+    if (c < H) {
+      T t[VEC];
+      *reinterpret_cast<uint4*>(t) = *reinterpret_cast<const uint4*>(x + base + c);
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        v[i][k] = DTraits<T>::to_f32(t[k]);
+        sum += v[i][k];
+        sumsq += v[i][k] * v[i][k];
+      }
     }
   }
-  sum = wave_reduce_sum(sum);
-  sumsq = wave_reduce_sum(sumsq);
+  block_reduce_sum2(sum, sumsq, red);
   const float mu = sum / H;
   const float var = fmaxf(sumsq / H - mu * mu, 0.f);
   const float rs = rsqrtf(var + eps);
-  if (lane == 0) {
+  if (threadIdx.x == 0) {
     mean[row] = mu;
     rstd[row] = rs;
   }
-  for (int c = lane * VEC; c < H; c += WAVE_SIZE * VEC) {
-    T v[VEC];
-    *reinterpret_cast<uint4*>(v) = *reinterpret_cast<const uint4*>(xr + c);
-    T o[VEC];
+#pragma unroll
+  for (int i = 0; i < ITEMS; ++i) {
+    const int c = cols[i];
+    if (c < H) {
+      T o[VEC];
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        o[k] = DTraits<T>::from_f32((v[i][k] - mu) * rs * gamma[c + k] +
+                                    beta[c + k]);
+      }
+      *reinterpret_cast<uint4*>(y + base + c) = *reinterpret_cast<const uint4*>(o);
+    }
+  }
+}
+
+// backward dx: one BLOCK per row, the mirror of ln_fwd_kernel — dy/x
+// loaded once into registers, s1/s2 via block reduce, dx written from
+// registers. dgamma/dbeta are NOT computed here (a block-per-row grid
+// cannot accumulate columns without atomics); see col_stats_kernel.
+template <typename T, int VEC, int ITEMS>
+__global__ void ln_bwd_dx_kernel(const T* __restrict__ dy,
+                                 const T* __restrict__ x,
+                                 const float* __restrict__ gamma,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ rstd,
+                                 T* __restrict__ dx, int rows, int H) {
+  __shared__ float red[32];
+  const int row = blockIdx.x;
+  if (row >= rows) return;
+  const int64_t base = static_cast<int64_t>(row) * H;
+  const float mu = mean[row], rs = rstd[row];
+
+  float dw[ITEMS][VEC], xh[ITEMS][VEC];
+  int cols[ITEMS];
+  float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+  for (int i = 0; i < ITEMS; ++i) {
+    const int c = (i * blockDim.x + threadIdx.x) * VEC;
+    cols[i] = c;
+    if (c < H) {
+      T dt[VEC], xt[VEC];
+      *reinterpret_cast<uint4*>(dt) = *reinterpret_cast<const uint4*>(dy + base + c);
+      *reinterpret_cast<uint4*>(xt) = *reinterpret_cast<const uint4*>(x + base + c);
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        xh[i][k] = (DTraits<T>::to_f32(xt[k]) - mu) * rs;
+        dw[i][k] = DTraits<T>::to_f32(dt[k]) * gamma[c + k];
+        s1 += dw[i][k] * xh[i][k];
+        s2 += dw[i][k];
+      }
+    }
+  }
+  block_reduce_sum2(s1, s2, red);
+  s1 /= H;
+  s2 /= H;
+#pragma unroll
+  for (int i = 0; i < ITEMS; ++i) {
+    const int c = cols[i];
+    if (c < H) {
+      T o[VEC];
+#pragma unroll
+      for (int k = 0; k < VEC; ++k)
+        o[k] = DTraits<T>::from_f32(rs * (dw[i][k] - s2 - xh[i][k] * s1));
+      *reinterpret_cast<uint4*>(dx + base + c) = *reinterpret_cast<const uint4*>(o);
+    }
+  }
+}
+
+// Column stats for the LN family backward: streaming column reduction
+//   dgamma[c] += dy[r][c] * (z[r][c]-mu)*rs,  dbeta[c] += dy[r][c]
+//   (+ dbias[c] += dx_src[r][c] when dx_src != nullptr)
+// Each thread owns VEC contiguous columns and loops its row chunk with
+// independent (software-pipelineable) loads; partials [n_chunks, nsl*H]
+// are folded by col_reduce_kernel. Shared by ln_bwd and bdrl_bwd.
+template <typename T, int VEC, bool HAS_BIAS>
+__global__ void col_stats_kernel(const T* __restrict__ dy,
+                                 const T* __restrict__ z,
+                                 const T* __restrict__ dx_src,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ rstd,
+                                 float* __restrict__ part, int rows, int H,
+                                 int rows_per_chunk) {
+  const int c = (blockIdx.x * blockDim.x + threadIdx.x) * VEC;
+  if (c >= H) return;
+  const int r0 = blockIdx.y * rows_per_chunk;
+  const int r1 = min(r0 + rows_per_chunk, rows);
+  float acc_g[VEC] = {}, acc_b[VEC] = {}, acc_bias[VEC] = {};
+  for (int r = r0; r < r1; ++r) {
+    const int64_t base = static_cast<int64_t>(r) * H + c;
+    const float mu = mean[r], rs = rstd[r];
+    T dt[VEC], zt[VEC], xt[VEC];
+    *reinterpret_cast<uint4*>(dt) = *reinterpret_cast<const uint4*>(dy + base);
+    *reinterpret_cast<uint4*>(zt) = *reinterpret_cast<const uint4*>(z + base);
+    if (HAS_BIAS)
+      *reinterpret_cast<uint4*>(xt) =
+          *reinterpret_cast<const uint4*>(dx_src + base);
 #pragma unroll
     for (int k = 0; k < VEC; ++k) {
-      float f = DTraits<T>::to_f32(v[k]);
-      o[k] = DTraits<T>::from_f32((f - mu) * rs * gamma[c + k] + beta[c + k]);
+      const float d = DTraits<T>::to_f32(dt[k]);
+      acc_g[k] += d * (DTraits<T>::to_f32(zt[k]) - mu) * rs;
+      acc_b[k] += d;
+      if (HAS_BIAS) acc_bias[k] += DTraits<T>::to_f32(xt[k]);
     }
-    *reinterpret_cast<uint4*>(yr + c) = *reinterpret_cast<const uint4*>(o);
+  }
+  const int nsl = HAS_BIAS ? 3 : 2;
+  float* p = part + static_cast<int64_t>(blockIdx.y) * nsl * H + c;
+#pragma unroll
+  for (int k = 0; k < VEC; ++k) {
+    p[k] = acc_g[k];
+    p[H + k] = acc_b[k];
+    if (HAS_BIAS) p[2 * H + k] = acc_bias[k];
   }
 }
 
-// backward: dx in one pass; dgamma/dbeta accumulate in LDS (fast LDS
-// f32 atomics) per block, one global write per column per block, then a
-// deterministic column-reduce kernel over the per-block partials.
-template <typename T, int VEC, int NW>
-__global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
-                              const float* __restrict__ gamma,
-                              const float* __restrict__ mean,
-                              const float* __restrict__ rstd,
-                              T* __restrict__ dx,
-                              float* __restrict__ part_dgamma,
-                              float* __restrict__ part_dbeta, int rows, int H,
-                              int rows_per_block) {
-  const int lane = threadIdx.x & (WAVE_SIZE - 1);
-  const int wave = threadIdx.x / WAVE_SIZE;
-  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  // per-wave private slabs: no atomics, no cross-wave contention
-  float* lg = reinterpret_cast<float*>(smem_raw) + wave * 2 * H;  // [NW][2H]
-  float* lb = lg + H;
-  for (int c = lane; c < 2 * H; c += WAVE_SIZE) lg[c] = 0.f;
-
-  const int row0 = blockIdx.x * rows_per_block;
-  const int row_end = min(row0 + rows_per_block, rows);
-  for (int r = row0 + wave; r < row_end; r += NW) {
-    const T* dyr = dy + static_cast<int64_t>(r) * H;
-    const T* xr = x + static_cast<int64_t>(r) * H;
-    T* dxr = dx + static_cast<int64_t>(r) * H;
-    const float mu = mean[r], rs = rstd[r];
-
-    float s1 = 0.f, s2 = 0.f;
-    for (int c = lane * VEC; c < H; c += WAVE_SIZE * VEC) {
-      T dv[VEC], xv[VEC];
-      *reinterpret_cast<uint4*>(dv) = *reinterpret_cast<const uint4*>(dyr + c);
-      *reinterpret_cast<uint4*>(xv) = *reinterpret_cast<const uint4*>(xr + c);
-#pragma unroll
-      for (int k = 0; k < VEC; ++k) {
-        float d = DTraits<T>::to_f32(dv[k]);
-        float xh = (DTraits<T>::to_f32(xv[k]) - mu) * rs;
-        float dw = d * gamma[c + k];
-        s1 += dw * xh;
-        s2 += dw;
-      }
-    }
-    s1 = wave_reduce_sum(s1) / H;
-    s2 = wave_reduce_sum(s2) / H;
-
-    for (int c = lane * VEC; c < H; c += WAVE_SIZE * VEC) {
-      T dv[VEC], xv[VEC], o[VEC];
-      *reinterpret_cast<uint4*>(dv) = *reinterpret_cast<const uint4*>(dyr + c);
-      *reinterpret_cast<uint4*>(xv) = *reinterpret_cast<const uint4*>(xr + c);
-#pragma unroll
-      for (int k = 0; k < VEC; ++k) {
-        float d = DTraits<T>::to_f32(dv[k]);
-        float xh = (DTraits<T>::to_f32(xv[k]) - mu) * rs;
-        float dw = d * gamma[c + k];
-        o[k] = DTraits<T>::from_f32(rs * (dw - s2 - xh * s1));
-        lg[c + k] += d * xh;  // this lane owns these columns in its slab
-        lb[c + k] += d;
-      }
-      *reinterpret_cast<uint4*>(dxr + c) = *reinterpret_cast<const uint4*>(o);
-    }
-  }
-  __syncthreads();
-  float* slab0 = reinterpret_cast<float*>(smem_raw);
-  float* pg = part_dgamma + static_cast<int64_t>(blockIdx.x) * H;
-  float* pb = part_dbeta + static_cast<int64_t>(blockIdx.x) * H;
-  for (int c = threadIdx.x; c < H; c += blockDim.x) {
-    float ag = 0.f, ab = 0.f;
-#pragma unroll
-    for (int w = 0; w < NW; ++w) {
-      ag += slab0[w * 2 * H + c];
-      ab += slab0[w * 2 * H + H + c];
-    }
-    pg[c] = ag;
-    pb[c] = ab;
+// instantiation helper shared with fused_residual.hip
+template <typename T, int VEC>
+void launch_col_stats(const T* dy, const T* z, const T* dx_src,
+                      const float* mean, const float* rstd, float* part,
+                      int rows, int H, int rows_per_chunk, int n_chunks,
+                      bool has_bias, hipStream_t stream) {
+  const int threads = tmin(256, ((H / VEC + 63) / 64) * 64);
+  dim3 grid((H / VEC + threads - 1) / threads, n_chunks);
+  if (has_bias) {
+    hipLaunchKernelGGL((col_stats_kernel<T, VEC, true>), grid, dim3(threads),
+                       0, stream, dy, z, dx_src, mean, rstd, part, rows, H,
+                       rows_per_chunk);
+  } else {
+    hipLaunchKernelGGL((col_stats_kernel<T, VEC, false>), grid, dim3(threads),
+                       0, stream, dy, z, dx_src, mean, rstd, part, rows, H,
+                       rows_per_chunk);
   }
 }
+
+template void launch_col_stats<__hip_bfloat16, 8>(
+    const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+    const float*, const float*, float*, int, int, int, int, bool, hipStream_t);
+template void launch_col_stats<__half, 8>(const __half*, const __half*,
+                                          const __half*, const float*,
+                                          const float*, float*, int, int, int,
+                                          int, bool, hipStream_t);
+template void launch_col_stats<float, 4>(const float*, const float*,
+                                         const float*, const float*,
+                                         const float*, float*, int, int, int,
+                                         int, bool, hipStream_t);
 
 // grid (ceil(H/256), ceil(nparts/kColChunk)); out must be zero-filled.
 __global__ void col_reduce_kernel(const float* __restrict__ parts, int nparts,
@@ -170,6 +239,23 @@ __global__ void col_reduce_kernel(const float* __restrict__ parts, int nparts,
     }                                                                        \
   }()
 
+// pick (threads, ITEMS) for the forward block-per-row shape
+template <typename LaunchFn>
+static void launch_by_items(int slices, LaunchFn&& fn) {
+  if (slices <= 256) {
+    fn(std::integral_constant<int, 1>{}, ((slices + 63) / 64) * 64);
+  } else if (slices <= 512) {
+    fn(std::integral_constant<int, 2>{}, 256);
+  } else if (slices <= 1024) {
+    fn(std::integral_constant<int, 2>{}, 512);
+  } else if (slices <= 2048) {
+    fn(std::integral_constant<int, 4>{}, 512);
+  } else {
+    TORCH_CHECK(slices <= 4096, "layernorm: H too large");
+    fn(std::integral_constant<int, 4>{}, 1024);
+  }
+}
+
 std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor gamma,
                                   torch::Tensor beta, double eps) {
   TORCH_CHECK(x.dim() == 2 && x.is_contiguous(), "ln_fwd: x must be 2D contiguous");
@@ -181,16 +267,17 @@ std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor gamma,
   auto mean = torch::empty({rows}, opts);
   auto rstd = torch::empty({rows}, opts);
   auto stream = at::hip::getCurrentHIPStream();
-  constexpr int WAVES = 4;
-  dim3 grid((rows + WAVES - 1) / WAVES), block(WAVES * WAVE_SIZE);
   DISPATCH_FLOATING(x.scalar_type(), "ln_fwd", [&] {
     TORCH_CHECK(H % kVec == 0, "ln_fwd: H must be a multiple of ", kVec);
-    hipLaunchKernelGGL((ln_fwd_kernel<scalar_t, kVec>), grid, block, 0, stream,
-                       reinterpret_cast<const scalar_t*>(x.data_ptr()),
-                       gamma_f.data_ptr<float>(), beta_f.data_ptr<float>(),
-                       reinterpret_cast<scalar_t*>(y.data_ptr()),
-                       mean.data_ptr<float>(), rstd.data_ptr<float>(), rows, H,
-                       static_cast<float>(eps));
+    launch_by_items(H / kVec, [&](auto items_c, int threads) {
+      hipLaunchKernelGGL((ln_fwd_kernel<scalar_t, kVec, decltype(items_c)::value>),
+                         dim3(rows), dim3(threads), 0, stream,
+                         reinterpret_cast<const scalar_t*>(x.data_ptr()),
+                         gamma_f.data_ptr<float>(), beta_f.data_ptr<float>(),
+                         reinterpret_cast<scalar_t*>(y.data_ptr()),
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(), rows,
+                         H, static_cast<float>(eps));
+    });
   });
   return {y, mean, rstd};
 }
@@ -202,42 +289,42 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x,
   const int rows = x.size(0), H = x.size(1);
   auto gamma_f = gamma.contiguous().to(torch::kFloat32);
   auto dx = torch::empty_like(x);
-  constexpr int NW = 4;                    // waves per block
-  const int rows_per_block = 16;           // 4 rows per wave
-  const int nblocks = (rows + rows_per_block - 1) / rows_per_block;
+  const int rows_per_chunk = 8;
+  const int n_chunks = (rows + rows_per_chunk - 1) / rows_per_chunk;
   auto opts = x.options().dtype(torch::kFloat32);
-  auto part_g = torch::empty({nblocks, H}, opts);
-  auto part_b = torch::empty({nblocks, H}, opts);
-  auto dgamma = torch::zeros({H}, opts);
-  auto dbeta = torch::zeros({H}, opts);
+  auto part = torch::empty({n_chunks, 2 * H}, opts);
+  auto dgb = torch::zeros({2 * H}, opts);
   auto stream = at::hip::getCurrentHIPStream();
-  dim3 grid(nblocks), block(NW * WAVE_SIZE);
-  const size_t lds = NW * 2 * static_cast<size_t>(H) * sizeof(float);
-  TORCH_CHECK(lds <= 160 * 1024, "ln_bwd: H too large for LDS accumulation");
   auto dy_c = dy.contiguous();
   DISPATCH_FLOATING(x.scalar_type(), "ln_bwd", [&] {
     TORCH_CHECK(H % kVec == 0, "ln_bwd: H must be a multiple of ", kVec);
-    hipLaunchKernelGGL((ln_bwd_kernel<scalar_t, kVec, NW>), grid,
-                       block, lds, stream,
-                       reinterpret_cast<const scalar_t*>(dy_c.data_ptr()),
-                       reinterpret_cast<const scalar_t*>(x.data_ptr()),
-                       gamma_f.data_ptr<float>(), mean.data_ptr<float>(),
-                       rstd.data_ptr<float>(),
-                       reinterpret_cast<scalar_t*>(dx.data_ptr()),
-                       part_g.data_ptr<float>(), part_b.data_ptr<float>(), rows,
-                       H, rows_per_block);
+    launch_by_items(H / kVec, [&](auto items_c, int threads) {
+      hipLaunchKernelGGL(
+          (ln_bwd_dx_kernel<scalar_t, kVec, decltype(items_c)::value>),
+          dim3(rows), dim3(threads), 0, stream,
+          reinterpret_cast<const scalar_t*>(dy_c.data_ptr()),
+          reinterpret_cast<const scalar_t*>(x.data_ptr()),
+          gamma_f.data_ptr<float>(), mean.data_ptr<float>(),
+          rstd.data_ptr<float>(),
+          reinterpret_cast<scalar_t*>(dx.data_ptr()), rows, H);
+    });
+    launch_col_stats<scalar_t, kVec>(
+        reinterpret_cast<const scalar_t*>(dy_c.data_ptr()),
+        reinterpret_cast<const scalar_t*>(x.data_ptr()), nullptr,
+        mean.data_ptr<float>(), rstd.data_ptr<float>(),
+        part.data_ptr<float>(), rows, H, rows_per_chunk, n_chunks, false,
+        stream);
   });
-  dim3 rgrid((H + 255) / 256, (nblocks + kColChunk - 1) / kColChunk), rblock(256);
-  hipLaunchKernelGGL(col_reduce_kernel, rgrid, rblock, 0, stream,
-                     part_g.data_ptr<float>(), nblocks, H,
-                     dgamma.data_ptr<float>());
-  hipLaunchKernelGGL(col_reduce_kernel, rgrid, rblock, 0, stream,
-                     part_b.data_ptr<float>(), nblocks, H,
-                     dbeta.data_ptr<float>());
+  dim3 rgrid((2 * H + 255) / 256, (n_chunks + kColChunk - 1) / kColChunk);
+  hipLaunchKernelGGL(col_reduce_kernel, rgrid, dim3(256), 0, stream,
+                     part.data_ptr<float>(), n_chunks, 2 * H,
+                     dgb.data_ptr<float>());
+  auto dgamma = dgb.narrow(0, 0, H);
+  auto dbeta = dgb.narrow(0, H, H);
   if (gamma.scalar_type() != torch::kFloat32) {
     return {dx, dgamma.to(gamma.scalar_type()), dbeta.to(gamma.scalar_type())};
   }
-  return {dx, dgamma, dbeta};
+  return {dx, dgamma.contiguous(), dbeta.contiguous()};
 }
 
 }  // namespace bpa
