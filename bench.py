@@ -58,6 +58,10 @@ def parse_args():
     p.add_argument("--local_batch", type=int, default=0)
     p.add_argument("--model_config", type=str,
                    default="config/bert_large_uncased_config.json")
+    p.add_argument("--pure_bf16", action="store_true",
+                   help="bf16 model weights + fp32 LAMB masters instead of "
+                        "fp32 weights + autocast (no per-microbatch weight "
+                        "casts; bf16 gradient all-reduce)")
     p.add_argument("--seed", type=int, default=1234)
     return p.parse_args()
 
@@ -100,6 +104,8 @@ def main():
     if config.vocab_size % 64:
         config.vocab_size += 64 - config.vocab_size % 64
     model = BertForPreTraining(config).to(device)
+    if args.pure_bf16:
+        model = model.to(torch.bfloat16)
     criterion = BertPretrainingCriterion(config.vocab_size)
     model = comm.wrap_ddp(model, local_rank)
     named = list(model.named_parameters())
@@ -112,6 +118,7 @@ def main():
              "weight_decay": 0.0},
         ],
         lr=phase["lr"],
+        master_weights=args.pure_bf16,
     )
     scheduler = PolyWarmUpScheduler(optimizer, warmup=0.2843, total_steps=7038)
     gen = torch.Generator(device=device).manual_seed(args.seed + rank)
@@ -121,7 +128,8 @@ def main():
         batch = make_batch(gen, device, bsz, seq, vocab_unpadded, phase["max_pred"])
         ids, tt, mask, labels, nsp = batch
         sync = (micro_idx + 1) % accum == 0
-        with torch.autocast(device.type, dtype=torch.bfloat16, enabled=use_cuda):
+        with torch.autocast(device.type, dtype=torch.bfloat16,
+                            enabled=use_cuda and not args.pure_bf16):
             scores, rel, glabels = model(ids, tt, mask, masked_lm_labels=labels)
             loss = criterion(scores, rel, glabels, nsp) / accum
         if sync or not isinstance(model, torch.nn.parallel.DistributedDataParallel):
@@ -179,7 +187,10 @@ def main():
                 "seq_len": seq,
                 "local_batch": bsz,
                 "accumulation": accum,
-                "optimizer": "FusedLAMB (HIP multi-tensor)",
+                "optimizer": "FusedLAMB (HIP multi-tensor)"
+                + (" + fp32 masters" if args.pure_bf16 else ""),
+                "weights": "bf16 (fp32 LAMB masters)" if args.pure_bf16
+                else "fp32 + bf16 autocast",
                 "parallelism": f"dp{world}",
             },
         }

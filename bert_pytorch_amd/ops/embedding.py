@@ -44,6 +44,7 @@ class _FusedEmbeddingLNDropout(torch.autograd.Function):
         ctx.max_pos = pos_emb.shape[0]
         ctx.n_types = tok_emb.shape[0] if tok_emb is not None else 0
         ctx.has_tok = tok_emb is not None
+        ctx.table_dtype = word_emb.dtype
         return y
 
     @staticmethod
@@ -54,6 +55,12 @@ class _FusedEmbeddingLNDropout(torch.autograd.Function):
             dy.contiguous(), ids, tt, z, mask, weight, mean, rstd, ctx.p,
             ctx.vocab, ctx.max_pos, ctx.n_types,
         )
+        if ctx.table_dtype != d_word.dtype:
+            # pure-bf16 mode: grads must match the low-precision tables
+            d_word = d_word.to(ctx.table_dtype)
+            d_pos = d_pos.to(ctx.table_dtype)
+            if ctx.has_tok:
+                d_tok = d_tok.to(ctx.table_dtype)
         return (
             None,
             None,

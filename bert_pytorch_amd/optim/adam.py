@@ -46,6 +46,12 @@ class FusedAdam(Optimizer):
             set_to_none = self.set_grad_none
         super().zero_grad(set_to_none=set_to_none)
 
+    def load_state_dict(self, state_dict):  # noqa: D102
+        from .lamb import _restore_fp32_state  # noqa: PLC0415
+
+        super().load_state_dict(state_dict)
+        _restore_fp32_state(self, state_dict, ("exp_avg", "exp_avg_sq"))
+
     @torch.no_grad()
     def step(self, closure=None):
         loss = None

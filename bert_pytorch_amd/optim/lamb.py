@@ -37,6 +37,25 @@ def _use_native(params: List[torch.Tensor]) -> bool:
     return True
 
 
+def _restore_fp32_state(optimizer, state_dict, keys) -> None:
+    """Re-load the named per-param state entries as fp32 (undoing the
+    param-dtype cast torch's Optimizer.load_state_dict applies)."""
+    id_map = {}
+    for saved_g, g in zip(state_dict["param_groups"], optimizer.param_groups):
+        for old_id, p in zip(saved_g["params"], g["params"]):
+            id_map[old_id] = p
+    for old_id, saved in state_dict["state"].items():
+        p = id_map.get(old_id)
+        if p is None or p.dtype == torch.float32:
+            continue
+        for k in keys:
+            t = saved.get(k)
+            if torch.is_tensor(t):
+                optimizer.state[p][k] = t.to(
+                    device=p.device, dtype=torch.float32
+                )
+
+
 class FusedLAMB(Optimizer):
     def __init__(
         self,
@@ -52,7 +71,14 @@ class FusedLAMB(Optimizer):
         set_grad_none: bool = True,
         max_grad_norm: float = 1.0,
         use_nvlamb: bool = False,
+        master_weights: bool = False,
     ):
+        """``master_weights=True`` keeps an fp32 master copy (plus a
+        persistent fp32 grad buffer) for every non-fp32 parameter: the
+        update runs entirely in fp32 and the bf16/fp16 parameter is the
+        cast-down of the master. This is the pure-bf16 training mode —
+        the model holds bf16 weights (no per-microbatch autocast weight
+        casts, bf16 gradient all-reduce) while LAMB math stays fp32."""
         if amsgrad:
             raise RuntimeError("FusedLAMB does not support amsgrad")
         defaults = dict(
@@ -68,11 +94,27 @@ class FusedLAMB(Optimizer):
         self.adam_w_mode = adam_w_mode
         self.set_grad_none = set_grad_none
         self.use_nvlamb = use_nvlamb
+        self.master_weights = master_weights
 
     def zero_grad(self, set_to_none: bool | None = None):  # noqa: D102
         if set_to_none is None:
             set_to_none = self.set_grad_none
         super().zero_grad(set_to_none=set_to_none)
+
+    def state_dict(self):  # noqa: D102
+        sd = super().state_dict()
+        for s in sd["state"].values():  # grad32 is per-step scratch
+            s.pop("grad32", None)
+        return sd
+
+    def load_state_dict(self, state_dict):  # noqa: D102
+        super().load_state_dict(state_dict)
+        # torch's Optimizer.load_state_dict casts floating state to the
+        # param dtype — which would silently truncate fp32 moments and
+        # masters to bf16 for low-precision params. Restore them from
+        # the ORIGINAL state_dict tensors.
+        _restore_fp32_state(self, state_dict,
+                            ("exp_avg", "exp_avg_sq", "master"))
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -94,37 +136,87 @@ class FusedLAMB(Optimizer):
             self._step_eager(all_grads)
         return loss
 
+    def _master_lists(self, params):
+        """(work_params, work_grads, cast_back) for one group: fp32
+        params pass through; low-precision params get fp32 master +
+        persistent fp32 grad buffers (batched _foreach_ casts)."""
+        work_p, work_g = [], []
+        lo_p, lo_masters, lo_g32, lo_grads = [], [], [], []
+        for p in params:
+            state = self.state[p]
+            if p.dtype == torch.float32:
+                work_p.append(p)
+                work_g.append(p.grad)
+                continue
+            if "master" not in state:
+                state["master"] = p.detach().float().contiguous()
+                state["grad32"] = torch.empty_like(state["master"])
+            lo_p.append(p)
+            lo_masters.append(state["master"])
+            lo_g32.append(state["grad32"])
+            lo_grads.append(p.grad)
+            work_p.append(state["master"])
+            work_g.append(state["grad32"])
+        if lo_g32:
+            torch._foreach_copy_(lo_g32, lo_grads)  # bf16 -> fp32 cast
+
+        def cast_back():
+            if lo_p:
+                torch._foreach_copy_(lo_p, lo_masters)  # fp32 -> bf16
+
+        return work_p, work_g, cast_back
+
     # -- native (HIP) ----------------------------------------------------
     def _step_native(self, all_grads):
         ext = ops.extension()
         max_grad_norm = self.param_groups[0]["max_grad_norm"]
-        gnorm_sq = ext.multi_tensor_l2norm_sq(all_grads)  # [1] f32 device
+
+        per_group = []
+        norm_grads = []
         for group in self.param_groups:
             params = [p for p in group["params"] if p.grad is not None]
             if not params:
                 continue
+            if self.master_weights:
+                work_p, work_g, cast_back = self._master_lists(params)
+            else:
+                work_p = params
+                work_g = [p.grad for p in params]
+                cast_back = None
+            per_group.append((group, work_p, work_g, cast_back))
+            norm_grads.extend(work_g)
+
+        gnorm_sq = ext.multi_tensor_l2norm_sq(norm_grads)  # [1] f32 device
+        for group, work_p, work_g, cast_back in per_group:
             group["step"] = group.get("step", 0) + 1
             step = group["step"]
-            grads, ms, vs = [], [], []
-            for p in params:
+            ms, vs = [], []
+            for p, wp in zip(
+                [p for p in group["params"] if p.grad is not None], work_p
+            ):
                 state = self.state[p]
-                if len(state) == 0:
-                    state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
-                    state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
+                if "exp_avg" not in state:
+                    state["exp_avg"] = torch.zeros_like(
+                        wp, dtype=torch.float32
+                    )
+                    state["exp_avg_sq"] = torch.zeros_like(
+                        wp, dtype=torch.float32
+                    )
                 state["step"] = step
-                grads.append(p.grad)
                 ms.append(state["exp_avg"])
                 vs.append(state["exp_avg_sq"])
             beta1, beta2 = group["betas"]
             wd = group["weight_decay"]
             use_ratio = self.use_nvlamb or wd != 0.0
             ext.fused_lamb(
-                params, grads, ms, vs, gnorm_sq,
+                work_p, work_g, ms, vs, gnorm_sq,
                 float(group["lr"]), beta1, beta2, group["eps"], wd,
                 step, bool(group["bias_correction"]),
                 bool(group["grad_averaging"]), float(max_grad_norm),
                 bool(use_ratio),
             )
+            if cast_back is not None:
+                cast_back()
 
     # -- eager reference -------------------------------------------------
     def _step_eager(self, all_grads):
@@ -149,21 +241,31 @@ class FusedLAMB(Optimizer):
             use_ratio = self.use_nvlamb or wd != 0.0
             for p in params:
                 state = self.state[p]
-                if len(state) == 0:
+                use_master = (
+                    self.master_weights and p.dtype != torch.float32
+                )
+                if "exp_avg" not in state:
                     state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
                     state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
+                    if use_master:
+                        state["master"] = p.detach().float()
                 state["step"] = step
+                w = state["master"] if use_master else p
                 g = p.grad.float() * clip_scale
                 m, v = state["exp_avg"], state["exp_avg_sq"]
                 m.mul_(beta1).add_(g, alpha=(1 - beta1) if group["grad_averaging"] else 1.0)
                 v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
                 update = (m / bc1) / ((v / bc2).sqrt() + group["eps"])
                 if wd != 0.0:
-                    update = update + wd * p.float()
+                    update = update + wd * w.float()
                 ratio = 1.0
                 if use_ratio:
-                    wn = float(p.float().norm())
+                    wn = float(w.float().norm())
                     un = float(update.norm())
                     if wn > 0 and un > 0:
                         ratio = wn / un
-                p.add_(update.to(p.dtype), alpha=-group["lr"] * ratio)
+                if use_master:
+                    w.add_(update, alpha=-group["lr"] * ratio)
+                    p.copy_(w.to(p.dtype))
+                else:
+                    p.add_(update.to(p.dtype), alpha=-group["lr"] * ratio)

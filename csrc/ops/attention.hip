@@ -371,6 +371,8 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
       sacc = MFMA16(frag_row(krow, 32, g), qfrag[1], sacc);
       dpacc = MFMA16(frag_row(vrow, 0, g), dofrag[0], dpacc);
       dpacc = MFMA16(frag_row(vrow, 32, g), dofrag[1], dpacc);
+      // one Philox call serves a consecutive key pair (key>>1 shared)
+      uint32_t r4[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int key = k0 + t * 16 + g * 4 + r;
@@ -379,10 +381,11 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
             valid ? __expf(sacc[r] * scale - lse_q) : 0.f;
         float dpd = dpacc[r];
         if (TRAIN_DROP) {
-          uint32_t r4[4];
-          philox(drop_base + static_cast<uint64_t>(q_row >> 1) * s2 +
-                     (key >> 1),
-                 r4);
+          if ((r & 1) == 0) {
+            philox(drop_base + static_cast<uint64_t>(q_row >> 1) * s2 +
+                       (key >> 1),
+                   r4);
+          }
           const bool keep =
               u32_to_uniform(r4[(q_row & 1) * 2 + (key & 1)]) >= p;
           dpd = keep ? dpd * inv_keep : 0.f;
@@ -555,6 +558,8 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       dp = MFMA16(frag_row(dorow, 0, g), frag_row(vrow, 0, g), dp);
       dp = MFMA16(frag_row(dorow, 32, g), frag_row(vrow, 32, g), dp);
       const bool kvalid = key_abs < slen && key_abs < S;
+      // one Philox call serves a consecutive q pair (q_abs>>1 shared)
+      uint32_t r4[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int q_abs = q0 + mq * 16 + g * 4 + r;
@@ -565,10 +570,11 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
                        : 0.f;
         float dpd = dp[r];
         if (TRAIN_DROP) {
-          uint32_t r4[4];
-          philox(drop_base + static_cast<uint64_t>(q_abs >> 1) * s2 +
-                     (key_abs >> 1),
-                 r4);
+          if ((r & 1) == 0) {
+            philox(drop_base + static_cast<uint64_t>(q_abs >> 1) * s2 +
+                       (key_abs >> 1),
+                   r4);
+          }
           const bool keep =
               u32_to_uniform(r4[(q_abs & 1) * 2 + (key_abs & 1)]) >= p;
           dpd = keep ? dpd * inv_keep : 0.f;

@@ -2,10 +2,12 @@
 //
 // BertEmbeddings in one kernel (reference: src/modeling.py:338-373):
 //   z = word[ids] + pos[s] (+ tok[tt]) ; y = dropout(LN(z))
-// Tables are fp32 parameters; the output is emitted in the autocast
-// dtype. Backward: dz from LN-backward (stored to a fp32 buffer), then
-// deterministic pos/token-type reductions and an atomic scatter-add
-// into the dense fp32 word-embedding gradient.
+// Tables may be fp32 (autocast mode) or bf16/fp16 (pure-bf16 master-
+// weight mode); the output is emitted in the requested dtype. Backward:
+// dz from LN-backward (stored to a fp32 buffer), then deterministic
+// pos/token-type reductions and an atomic scatter-add into a dense fp32
+// word-embedding gradient (cast to the table dtype by the Python
+// wrapper when the tables are low-precision).
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -17,76 +19,102 @@ namespace bpa {
 __global__ void col_reduce_kernel(const float* __restrict__ parts, int nparts,
                                   int H, float* __restrict__ out);
 
-template <typename T, int VEC, bool HAS_TOK, bool TRAIN_DROP>
+template <typename T, typename TW, int VEC, bool HAS_TOK, bool TRAIN_DROP>
 __global__ void embed_fwd_kernel(
     const int64_t* __restrict__ ids, const int64_t* __restrict__ tt,
-    const float* __restrict__ word, const float* __restrict__ pos,
-    const float* __restrict__ tok, const float* __restrict__ gamma,
+    const TW* __restrict__ word, const TW* __restrict__ pos,
+    const TW* __restrict__ tok, const float* __restrict__ gamma,
     const float* __restrict__ beta, T* __restrict__ y, float* __restrict__ z,
     uint8_t* __restrict__ mask, float* __restrict__ mean,
     float* __restrict__ rstd, int rows, int seq_len, int H, float p, float eps,
     uint64_t seed, uint64_t offset) {
-  const int lane = threadIdx.x & (WAVE_SIZE - 1);
-  const int wave = threadIdx.x / WAVE_SIZE;
-  const int row = blockIdx.x * (blockDim.x / WAVE_SIZE) + wave;
+  // one BLOCK per row; ITEMS elements per thread held in registers
+  // through the mean/var reduction (see layernorm.hip ln_fwd_kernel)
+  __shared__ float red[32];
+  const int row = blockIdx.x;
   if (row >= rows) return;
   const int64_t base = static_cast<int64_t>(row) * H;
   const int s = row % seq_len;
   const int64_t wid = ids[row];
-  const int64_t tid = HAS_TOK ? tt[row] : 0;
-  const float* wrow = word + wid * H;
-  const float* prow = pos + static_cast<int64_t>(s) * H;
-  const float* trow = HAS_TOK ? tok + tid * H : nullptr;
+  const int64_t tid_ = HAS_TOK ? tt[row] : 0;
+  const TW* wrow = word + wid * H;
+  const TW* prow = pos + static_cast<int64_t>(s) * H;
+  const TW* trow = HAS_TOK ? tok + tid_ * H : nullptr;
   const float keep_scale = TRAIN_DROP ? 1.0f / (1.0f - p) : 1.0f;
   Philox philox(seed);
 
+  constexpr int ITEMS = 2;  // covers H <= 2*blockDim*VEC
+  float zv[ITEMS][VEC];
+  int cols[ITEMS];
   float sum = 0.f, sumsq = 0.f;
-  for (int c = lane * VEC; c < H; c += WAVE_SIZE * VEC) {
 #pragma unroll
-    for (int k = 0; k < VEC; ++k) {
-      float t = wrow[c + k] + prow[c + k];
-      if (HAS_TOK) t += trow[c + k];
-      z[base + c + k] = t;
-      sum += t;
-      sumsq += t * t;
+  for (int i = 0; i < ITEMS; ++i) {
+    const int c = (i * blockDim.x + threadIdx.x) * VEC;
+    cols[i] = c;
+    if (c < H) {
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        float t = DTraits<TW>::to_f32(wrow[c + k]) +
+                  DTraits<TW>::to_f32(prow[c + k]);
+        if (HAS_TOK) t += DTraits<TW>::to_f32(trow[c + k]);
+        zv[i][k] = t;
+        sum += t;
+        sumsq += t * t;
+      }
+      if (sizeof(float) * VEC == 16) {
+        *reinterpret_cast<uint4*>(z + base + c) =
+            *reinterpret_cast<const uint4*>(zv[i]);
+      } else {
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) z[base + c + k] = zv[i][k];
+      }
     }
   }
-  sum = wave_reduce_sum(sum);
-  sumsq = wave_reduce_sum(sumsq);
+  block_reduce_sum2(sum, sumsq, red);
   const float mu = sum / H;
   const float var = fmaxf(sumsq / H - mu * mu, 0.f);
   const float rs = rsqrtf(var + eps);
-  if (lane == 0) {
+  if (threadIdx.x == 0) {
     mean[row] = mu;
     rstd[row] = rs;
   }
-  for (int c = lane * VEC; c < H; c += WAVE_SIZE * VEC) {
-    T ov[VEC];
-    uint8_t mv[VEC];
-    if (TRAIN_DROP) {
 #pragma unroll
-      for (int q = 0; q < VEC / 4; ++q) {
-        uint32_t r4[4];
-        philox(offset + (base + c) / 4 + q, r4);
+  for (int i = 0; i < ITEMS; ++i) {
+    const int c = cols[i];
+    if (c < H) {
+      T ov[VEC];
+      uint8_t mv[VEC];
+      if (TRAIN_DROP) {
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
-          mv[q * 4 + j] = u32_to_uniform(r4[j]) >= p ? 1 : 0;
+        for (int q = 0; q < VEC / 4; ++q) {
+          uint32_t r4[4];
+          philox(offset + (base + c) / 4 + q, r4);
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            mv[q * 4 + j] = u32_to_uniform(r4[j]) >= p ? 1 : 0;
+        }
       }
-    }
 #pragma unroll
-    for (int k = 0; k < VEC; ++k) {
-      float t = (z[base + c + k] - mu) * rs * gamma[c + k] + beta[c + k];
-      if (TRAIN_DROP) t = mv[k] ? t * keep_scale : 0.f;
-      ov[k] = DTraits<T>::from_f32(t);
-    }
-    *reinterpret_cast<uint4*>(y + base + c) = *reinterpret_cast<const uint4*>(ov);
-    if (TRAIN_DROP) {
-      if (VEC == 8)
-        *reinterpret_cast<uint2*>(mask + base + c) =
-            *reinterpret_cast<const uint2*>(mv);
-      else
-        *reinterpret_cast<uint32_t*>(mask + base + c) =
-            *reinterpret_cast<const uint32_t*>(mv);
+      for (int k = 0; k < VEC; ++k) {
+        float t = (zv[i][k] - mu) * rs * gamma[c + k] + beta[c + k];
+        if (TRAIN_DROP) t = mv[k] ? t * keep_scale : 0.f;
+        ov[k] = DTraits<T>::from_f32(t);
+      }
+      if (sizeof(T) * VEC == 16) {
+        *reinterpret_cast<uint4*>(y + base + c) =
+            *reinterpret_cast<const uint4*>(ov);
+      } else {
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) y[base + c + k] = ov[k];
+      }
+      if (TRAIN_DROP) {
+        if (VEC == 8)
+          *reinterpret_cast<uint2*>(mask + base + c) =
+              *reinterpret_cast<const uint2*>(mv);
+        else
+          *reinterpret_cast<uint32_t*>(mask + base + c) =
+              *reinterpret_cast<const uint32_t*>(mv);
+      }
     }
   }
 }
@@ -243,8 +271,10 @@ std::vector<torch::Tensor> embedding_ln_dropout_fwd(
     torch::Tensor beta, double p, double eps, int64_t seed, int64_t offset,
     torch::ScalarType out_dtype) {
   TORCH_CHECK(ids.dim() == 2, "embed_fwd: ids must be [B, S]");
-  TORCH_CHECK(word.scalar_type() == torch::kFloat32,
-              "embed_fwd: tables must be fp32 master params");
+  TORCH_CHECK(pos.scalar_type() == word.scalar_type() &&
+                  (!tok.has_value() ||
+                   tok->scalar_type() == word.scalar_type()),
+              "embed_fwd: all tables must share a dtype");
   const int batch = ids.size(0), seq_len = ids.size(1);
   const int rows = batch * seq_len;
   const int H = word.size(1);
@@ -258,7 +288,7 @@ std::vector<torch::Tensor> embedding_ln_dropout_fwd(
 
   auto y = torch::empty({batch, seq_len, H},
                         word.options().dtype(out_dtype));
-  auto fopts = word.options();
+  auto fopts = word.options().dtype(torch::kFloat32);
   auto z = torch::empty({rows, H}, fopts);
   auto mask = train_drop
                   ? torch::empty({rows, H}, word.options().dtype(torch::kUInt8))
@@ -266,29 +296,41 @@ std::vector<torch::Tensor> embedding_ln_dropout_fwd(
   auto mean = torch::empty({rows}, fopts);
   auto rstd = torch::empty({rows}, fopts);
   auto stream = at::hip::getCurrentHIPStream();
-  constexpr int WAVES = 4;
-  dim3 grid((rows + WAVES - 1) / WAVES), block(WAVES * WAVE_SIZE);
   DISPATCH_OUT(out_dtype, "embed_fwd", [&] {
+    using scalar_out_t = out_t;
+    constexpr int kVecOuter = kVec;
     TORCH_CHECK(H % kVec == 0, "embed_fwd: H % ", kVec, " != 0");
-    auto launch = [&](auto tok_c, auto train_c) {
-      hipLaunchKernelGGL(
-          (embed_fwd_kernel<out_t, kVec, decltype(tok_c)::value,
-                            decltype(train_c)::value>),
-          grid, block, 0, stream, ids_c.data_ptr<int64_t>(),
-          has_tok ? tt_c.data_ptr<int64_t>() : nullptr,
-          word.data_ptr<float>(), pos.data_ptr<float>(),
-          has_tok ? tok->data_ptr<float>() : nullptr,
-          gamma_f.data_ptr<float>(), beta_f.data_ptr<float>(),
-          reinterpret_cast<out_t*>(y.data_ptr()), z.data_ptr<float>(),
-          train_drop ? mask.data_ptr<uint8_t>() : nullptr,
-          mean.data_ptr<float>(), rstd.data_ptr<float>(), rows, seq_len, H,
-          static_cast<float>(p), static_cast<float>(eps),
-          static_cast<uint64_t>(seed), static_cast<uint64_t>(offset));
-    };
-    if (has_tok && train_drop) launch(std::true_type{}, std::true_type{});
-    else if (has_tok) launch(std::true_type{}, std::false_type{});
-    else if (train_drop) launch(std::false_type{}, std::true_type{});
-    else launch(std::false_type{}, std::false_type{});
+    const int slices = H / kVec;
+    const int threads =
+        tmin(1024, (((slices + 1) / 2 + WAVE_SIZE - 1) / WAVE_SIZE) *
+                       WAVE_SIZE);
+    TORCH_CHECK(slices <= 2 * threads, "embed_fwd: H too large");
+    DISPATCH_OUT(word.scalar_type(), "embed_fwd_table", [&] {
+      using tw_t = out_t;  // inner dispatch alias
+      auto launch = [&](auto tok_flag, auto train_c) {
+        hipLaunchKernelGGL(
+            (embed_fwd_kernel<scalar_out_t, tw_t, kVecOuter,
+                              decltype(tok_flag)::value,
+                              decltype(train_c)::value>),
+            dim3(rows), dim3(threads), 0, stream, ids_c.data_ptr<int64_t>(),
+            has_tok ? tt_c.data_ptr<int64_t>() : nullptr,
+            reinterpret_cast<const tw_t*>(word.data_ptr()),
+            reinterpret_cast<const tw_t*>(pos.data_ptr()),
+            has_tok ? reinterpret_cast<const tw_t*>(tok->data_ptr())
+                    : nullptr,
+            gamma_f.data_ptr<float>(), beta_f.data_ptr<float>(),
+            reinterpret_cast<scalar_out_t*>(y.data_ptr()),
+            z.data_ptr<float>(),
+            train_drop ? mask.data_ptr<uint8_t>() : nullptr,
+            mean.data_ptr<float>(), rstd.data_ptr<float>(), rows, seq_len,
+            H, static_cast<float>(p), static_cast<float>(eps),
+            static_cast<uint64_t>(seed), static_cast<uint64_t>(offset));
+      };
+      if (has_tok && train_drop) launch(std::true_type{}, std::true_type{});
+      else if (has_tok) launch(std::true_type{}, std::false_type{});
+      else if (train_drop) launch(std::false_type{}, std::true_type{});
+      else launch(std::false_type{}, std::false_type{});
+    });
   });
   return {y, z, mask, mean, rstd};
 }

@@ -178,3 +178,34 @@ def test_optimizer_state_dict_roundtrip():
     opt2.load_state_dict(state)
     # the group 'step' key survives (required by Poly/Linear schedulers)
     assert opt2.param_groups[0]["step"] == 3
+
+
+def test_fused_lamb_master_weights_bf16():
+    """bf16 params + fp32 masters: loss decreases, master holds the
+    precise weights, params are the cast-down of the master."""
+    import torch
+    from bert_pytorch_amd.optim import FusedLAMB
+
+    torch.manual_seed(0)
+    lin = torch.nn.Linear(16, 16).to(torch.bfloat16)
+    opt = FusedLAMB(lin.parameters(), lr=1e-2, master_weights=True)
+    x = torch.randn(32, 16).to(torch.bfloat16)
+    losses = []
+    for _ in range(6):
+        opt.zero_grad()
+        loss = (lin(x) ** 2).float().mean()
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0]
+    state = opt.state[lin.weight]
+    assert state["master"].dtype == torch.float32
+    assert torch.equal(lin.weight.data,
+                       state["master"].to(torch.bfloat16))
+    # state_dict round-trip preserves the master copy
+    sd = opt.state_dict()
+    lin2 = torch.nn.Linear(16, 16).to(torch.bfloat16)
+    opt2 = FusedLAMB(lin2.parameters(), lr=1e-2, master_weights=True)
+    opt2.load_state_dict(sd)
+    st2 = next(iter(opt2.state.values()))
+    assert "master" in st2 and st2["master"].dtype == torch.float32
