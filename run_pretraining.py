@@ -63,6 +63,13 @@ def parse_arguments(args=None) -> argparse.Namespace:
     parser.add_argument("--warmup_proportion", type=float, default=0.2843)
     parser.add_argument("--lr_decay", type=str, default="poly",
                         choices=["poly", "linear", "cosine", "constant"])
+    parser.add_argument("--optimizer", type=str, default="lamb",
+                        choices=["lamb", "adam"],
+                        help="fused LAMB (BERT two-phase) or fused Adam "
+                             "(RoBERTa single-phase)")
+    parser.add_argument("--pure_bf16", action="store_true",
+                        help="bf16 model weights + fp32 optimizer masters "
+                             "instead of fp32 weights + autocast")
     parser.add_argument("--seed", type=int, default=42)
     parser.add_argument("--fp16", action="store_true",
                         help="fp16 autocast + dynamic GradScaler")
@@ -163,6 +170,8 @@ def prepare_model(args, device):
                 strict=False,
             )
     model.to(device)
+    if args.pure_bf16:
+        model.to(torch.bfloat16)
     if args.checkpoint_activations:
         model.checkpoint_activations(True)
     global_steps = max(0, resume_step - args.previous_phase_end_step)
@@ -175,14 +184,33 @@ def prepare_optimizers(args, model, resume_state):
     no_decay = ["bias", "LayerNorm.weight", "LayerNorm.bias", "qkv_bias"]
     decay_params = [p for n, p in named if not any(d in n for d in no_decay)]
     nodecay_params = [p for n, p in named if any(d in n for d in no_decay)]
-    optimizer = FusedLAMB(
-        [
-            {"params": decay_params, "weight_decay": 0.01},
-            {"params": nodecay_params, "weight_decay": 0.0},
-        ],
-        lr=args.learning_rate,
+    groups = [
+        {"params": decay_params, "weight_decay": 0.01},
+        {"params": nodecay_params, "weight_decay": 0.0},
+    ]
+    if args.optimizer == "adam":  # RoBERTa path (BASELINE config 4)
+        from bert_pytorch_amd.optim import FusedAdam  # noqa: PLC0415
+
+        optimizer = FusedAdam(
+            groups, lr=args.learning_rate, master_weights=args.pure_bf16
+        )
+    else:
+        optimizer = FusedLAMB(
+            groups, lr=args.learning_rate, master_weights=args.pure_bf16
+        )
+    from bert_pytorch_amd.optim import (  # noqa: PLC0415
+        ConstantWarmUpScheduler,
+        CosineWarmUpScheduler,
+        LinearWarmUpScheduler,
     )
-    scheduler = PolyWarmUpScheduler(
+
+    sched_cls = {
+        "poly": PolyWarmUpScheduler,
+        "linear": LinearWarmUpScheduler,
+        "cosine": CosineWarmUpScheduler,
+        "constant": ConstantWarmUpScheduler,
+    }[args.lr_decay]
+    scheduler = sched_cls(
         optimizer, warmup=args.warmup_proportion, total_steps=args.max_steps
     )
     scaler = torch.amp.GradScaler("cuda", enabled=args.fp16 and not args.bf16)
@@ -328,7 +356,9 @@ def main(args) -> int:
     loader, sampler = prepare_dataset(args, resume_state)
 
     autocast_dtype = None
-    if args.bf16:
+    if args.pure_bf16:
+        autocast_dtype = None  # weights already bf16; no autocast casts
+    elif args.bf16:
         autocast_dtype = torch.bfloat16
     elif args.fp16:
         autocast_dtype = torch.float16

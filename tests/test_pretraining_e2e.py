@@ -122,3 +122,52 @@ def test_config_file_merge(workspace, tmp_path):
     assert args.max_steps == 9  # CLI wins
     assert args.local_batch_size == 2
     assert args.model_config_file == cfg_path
+
+
+def test_roberta_path_adam_no_nsp(tmp_path):
+    """RoBERTa single-phase: no NSP (2-entry special_token_positions),
+    fused Adam, linear decay (BASELINE config 4 semantics on CPU)."""
+    data_dir = tmp_path / "data"
+    synth.make_dataset(
+        str(data_dir), num_shards=1, samples_per_shard=32, seq_len=32,
+        vocab_size=512, seed=0, nsp=False,
+    )
+    model_cfg = {
+        "vocab_size": 512, "hidden_size": 64, "num_hidden_layers": 2,
+        "num_attention_heads": 4, "intermediate_size": 128,
+        "max_position_embeddings": 64, "type_vocab_size": 2,
+        "hidden_act": "gelu", "hidden_dropout_prob": 0.1,
+        "attention_probs_dropout_prob": 0.1, "initializer_range": 0.02,
+        "next_sentence": False,
+    }
+    cfg_path = tmp_path / "roberta.json"
+    cfg_path.write_text(json.dumps(model_cfg))
+    args = _args(
+        tmp_path, str(data_dir), str(cfg_path),
+        optimizer="adam", lr_decay="linear", max_steps="4",
+    )
+    steps = run_pretraining.main(args)
+    assert steps == 4
+    ckpts = list((tmp_path / "out" / "pretrain_ckpts").glob("ckpt_*.pt"))
+    assert ckpts
+    state = torch.load(ckpts[0], map_location="cpu", weights_only=False)
+    # NSP head absent from the checkpoint for the RoBERTa config
+    assert not any("seq_relationship" in k for k in state["model"])
+
+
+def test_pure_bf16_mode(workspace):
+    """--pure_bf16: bf16 weights with fp32 masters trains and checkpoints."""
+    tmp_path, data_dir, cfg_path = workspace
+    args = _args(tmp_path, data_dir, cfg_path, max_steps="3")
+    args.pure_bf16 = True
+    steps = run_pretraining.main(args)
+    assert steps == 3
+    ckpts = sorted((tmp_path / "out" / "pretrain_ckpts").glob("ckpt_*.pt"))
+    state = torch.load(ckpts[-1], map_location="cpu", weights_only=False)
+    w = next(v for k, v in state["model"].items() if "word_embeddings" in k)
+    assert w.dtype == torch.bfloat16
+    # optimizer state keeps fp32 masters
+    assert any(
+        "master" in s
+        for s in state["optimizer"]["state"].values()
+    )
