@@ -120,7 +120,7 @@ class _PackedQKV(nn.Module):
         self.qkv_bias = nn.Parameter(torch.zeros(3 * hidden_size))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return F.linear(x, self.qkv_weight, self.qkv_bias)
+        return ops.fused_linear(x, self.qkv_weight, self.qkv_bias)
 
     def _save_to_state_dict(self, destination, prefix, keep_vars):
         h = self.hidden_size

@@ -11,6 +11,7 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x,
 torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias);
 std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor bias);
+torch::Tensor col_sum(torch::Tensor x);
 std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
     torch::Tensor x, c10::optional<torch::Tensor> bias, torch::Tensor residual,
     torch::Tensor gamma, torch::Tensor beta, double p, double eps,
@@ -77,6 +78,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ln_bwd", &bpa::ln_bwd, "fused LayerNorm backward");
   m.def("bias_gelu_fwd", &bpa::bias_gelu_fwd);
   m.def("bias_gelu_bwd", &bpa::bias_gelu_bwd);
+  m.def("col_sum", &bpa::col_sum, "column sum [rows,H] -> fp32 [H]");
   m.def("bias_dropout_residual_ln_fwd", &bpa::bias_dropout_residual_ln_fwd);
   m.def("bias_dropout_residual_ln_bwd", &bpa::bias_dropout_residual_ln_bwd);
   m.def("embedding_ln_dropout_fwd", &bpa::embedding_ln_dropout_fwd);

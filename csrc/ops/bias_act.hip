@@ -113,6 +113,30 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
   }
 }
 
+// Generic column sum [rows, H] -> fp32 [H] (two-stage, deterministic).
+// Used for the packed-QKV bias gradient (replacing ATen's bf16 reduce
+// at ~2.5 TB/s with a streaming partial pass + col_reduce).
+template <typename T, int VEC>
+__global__ void col_sum_partial_kernel(const T* __restrict__ x,
+                                       float* __restrict__ part, int rows,
+                                       int H, int rows_per_chunk) {
+  const int c = (blockIdx.x * blockDim.x + threadIdx.x) * VEC;
+  if (c >= H) return;
+  const int r0 = blockIdx.y * rows_per_chunk;
+  const int r1 = min(r0 + rows_per_chunk, rows);
+  float acc[VEC] = {};
+  for (int r = r0; r < r1; ++r) {
+    T v[VEC];
+    *reinterpret_cast<uint4*>(v) =
+        *reinterpret_cast<const uint4*>(x + static_cast<int64_t>(r) * H + c);
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) acc[k] += DTraits<T>::to_f32(v[k]);
+  }
+  float* p = part + static_cast<int64_t>(blockIdx.y) * H + c;
+#pragma unroll
+  for (int k = 0; k < VEC; ++k) p[k] = acc[k];
+}
+
 #define DISPATCH_FLOATING2(TYPE, NAME, ...)                                  \
   [&] {                                                                      \
     if (TYPE == at::kBFloat16) {                                             \
@@ -167,7 +191,7 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
   const int rows_per_block = 16;
   const int stripe_count =
       static_cast<int>((rows + rows_per_block - 1) / rows_per_block);
-  auto dbias = torch::zeros({H}, fopts);
+  auto dbias = torch::empty({H}, fopts);
   auto dy_c = dy.contiguous();
   DISPATCH_FLOATING2(x.scalar_type(), "bias_gelu_bwd", [&] {
     TORCH_CHECK(H % kVec == 0, "bias_gelu_bwd: H % ", kVec, " != 0");
@@ -183,8 +207,7 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
                        reinterpret_cast<scalar_t*>(dx.data_ptr()),
                        part.data_ptr<float>(), static_cast<int>(rows), H,
                        rows_per_block);
-    dim3 rgrid((H + 255) / 256,
-               (stripe_count + kColChunk - 1) / kColChunk);
+    dim3 rgrid((H + 255) / 256);
     hipLaunchKernelGGL(col_reduce_kernel, rgrid, dim3(256), 0, stream,
                        part.data_ptr<float>(), stripe_count, H,
                        dbias.data_ptr<float>());
@@ -193,6 +216,32 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
     return {dx, dbias.to(bias.scalar_type())};
   }
   return {dx, dbias};
+}
+
+torch::Tensor col_sum(torch::Tensor x) {
+  TORCH_CHECK(x.dim() == 2 && x.is_contiguous(), "col_sum: bad x");
+  const int rows = x.size(0);
+  const int H = x.size(1);
+  const int rows_per_chunk = 32;
+  const int n_chunks = (rows + rows_per_chunk - 1) / rows_per_chunk;
+  auto fopts = x.options().dtype(torch::kFloat32);
+  auto out = torch::empty({H}, fopts);
+  auto stream = at::hip::getCurrentHIPStream();
+  DISPATCH_FLOATING2(x.scalar_type(), "col_sum", [&] {
+    TORCH_CHECK(H % kVec == 0, "col_sum: H % ", kVec, " != 0");
+    auto part = torch::empty({n_chunks, H}, fopts);
+    const int slices = H / kVec;
+    const int threads = tmin(256, ((slices + 63) / 64) * 64);
+    dim3 grid((slices + threads - 1) / threads, n_chunks);
+    hipLaunchKernelGGL((col_sum_partial_kernel<scalar_t, kVec>), grid,
+                       dim3(threads), 0, stream,
+                       reinterpret_cast<const scalar_t*>(x.data_ptr()),
+                       part.data_ptr<float>(), rows, H, rows_per_chunk);
+    hipLaunchKernelGGL(col_reduce_kernel, dim3((H + 255) / 256), dim3(256),
+                       0, stream, part.data_ptr<float>(), n_chunks, H,
+                       out.data_ptr<float>());
+  });
+  return out;
 }
 
 }  // namespace bpa

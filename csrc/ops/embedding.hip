@@ -382,13 +382,13 @@ std::vector<torch::Tensor> embedding_ln_dropout_bwd(
   });
 
   auto d_word = torch::zeros({vocab, H}, fopts);
-  auto d_pos = torch::zeros({max_pos, H}, fopts);
-  auto d_tok = has_tok ? torch::zeros({n_types, H}, fopts)
+  auto d_pos = torch::empty({max_pos, H}, fopts);
+  auto d_tok = has_tok ? torch::empty({n_types, H}, fopts)
                        : torch::empty({0}, fopts);
-  auto dgamma = torch::zeros({H}, fopts);
-  auto dbeta = torch::zeros({H}, fopts);
+  auto dgamma = torch::empty({H}, fopts);
+  auto dbeta = torch::empty({H}, fopts);
 
-  dim3 rgrid((H + 255) / 256, (nblocks + kColChunk - 1) / kColChunk), rblock(256);
+  dim3 rgrid((H + 255) / 256), rblock(256);
   hipLaunchKernelGGL(col_reduce_kernel, rgrid, rblock, 0, stream,
                      part_g.data_ptr<float>(), nblocks, H,
                      dgamma.data_ptr<float>());
@@ -415,9 +415,7 @@ std::vector<torch::Tensor> embedding_ln_dropout_bwd(
                        dz.data_ptr<float>(), tt_c.data_ptr<int64_t>(),
                        tparts.data_ptr<float>(), rows, H,
                        static_cast<int>(n_types), rows_per_block);
-    dim3 tgrid((n_types * H + 255) / 256,
-               (nblocks + kColChunk - 1) / kColChunk);
-    d_tok.zero_();
+    dim3 tgrid((n_types * H + 255) / 256);
     hipLaunchKernelGGL(col_reduce_kernel, tgrid, rblock, 0, stream,
                        tparts.data_ptr<float>(), nblocks,
                        static_cast<int>(n_types * H),

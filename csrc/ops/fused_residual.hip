@@ -290,7 +290,7 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
   const int n_slabs = has_bias ? 3 : 2;
   auto fopts = z.options().dtype(torch::kFloat32);
   auto part = torch::empty({n_chunks, n_slabs * H}, fopts);
-  auto out = torch::zeros({n_slabs * H}, fopts);
+  auto out = torch::empty({n_slabs * H}, fopts);
   const bool train_drop = p > 0.0;
   auto stream = at::hip::getCurrentHIPStream();
   DISPATCH_T(z.scalar_type(), "bdrl_bwd", [&] {
@@ -337,8 +337,7 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
         part.data_ptr<float>(), rows, H, rows_per_chunk, n_chunks, has_bias,
         stream);
   });
-  dim3 rgrid((n_slabs * H + 255) / 256,
-             (n_chunks + kColChunk - 1) / kColChunk);
+  dim3 rgrid((n_slabs * H + 255) / 256);
   hipLaunchKernelGGL(col_reduce_kernel, rgrid, dim3(256), 0, stream,
                      part.data_ptr<float>(), n_chunks, n_slabs * H,
                      out.data_ptr<float>());

@@ -207,17 +207,23 @@ template void launch_col_stats<float, 4>(const float*, const float*,
                                          const float*, float*, int, int, int,
                                          int, bool, hipStream_t);
 
-// grid (ceil(H/256), ceil(nparts/kColChunk)); out must be zero-filled.
+// Deterministic single-pass column reduce: grid (ceil(H/256)); each
+// thread owns one output column and streams all `nparts` partial rows
+// (coalesced across lanes, loads pipeline across iterations). Direct
+// store - no atomics, so `out` needs NO zero-fill (the old atomic
+// version forced a zeros() kernel launch per call site).
 __global__ void col_reduce_kernel(const float* __restrict__ parts, int nparts,
                                   int H, float* __restrict__ out) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= H) return;
-  const int p0 = blockIdx.y * kColChunk;
-  const int p1 = min(p0 + kColChunk, nparts);
   float acc = 0.f;
-  for (int p = p0; p < p1; ++p)
-    acc += parts[static_cast<int64_t>(p) * H + c];
-  atomicAdd(&out[c], acc);
+  int p = 0;
+  for (; p + 4 <= nparts; p += 4) {
+    const int64_t b = static_cast<int64_t>(p) * H + c;
+    acc += parts[b] + parts[b + H] + parts[b + 2 * H] + parts[b + 3 * H];
+  }
+  for (; p < nparts; ++p) acc += parts[static_cast<int64_t>(p) * H + c];
+  out[c] = acc;
 }
 
 #define DISPATCH_FLOATING(TYPE, NAME, ...)                                   \
@@ -293,7 +299,7 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x,
   const int n_chunks = (rows + rows_per_chunk - 1) / rows_per_chunk;
   auto opts = x.options().dtype(torch::kFloat32);
   auto part = torch::empty({n_chunks, 2 * H}, opts);
-  auto dgb = torch::zeros({2 * H}, opts);
+  auto dgb = torch::empty({2 * H}, opts);
   auto stream = at::hip::getCurrentHIPStream();
   auto dy_c = dy.contiguous();
   DISPATCH_FLOATING(x.scalar_type(), "ln_bwd", [&] {
@@ -315,7 +321,7 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x,
         part.data_ptr<float>(), rows, H, rows_per_chunk, n_chunks, false,
         stream);
   });
-  dim3 rgrid((2 * H + 255) / 256, (n_chunks + kColChunk - 1) / kColChunk);
+  dim3 rgrid((2 * H + 255) / 256);
   hipLaunchKernelGGL(col_reduce_kernel, rgrid, dim3(256), 0, stream,
                      part.data_ptr<float>(), n_chunks, 2 * H,
                      dgb.data_ptr<float>());

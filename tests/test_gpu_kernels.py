@@ -420,3 +420,52 @@ def test_gpu_matches_cpu_forward():
     with torch.no_grad(), torch.autocast("cuda", dtype=torch.bfloat16):
         s_gpu, r_gpu = model_gpu(ids.to(DEV), tt.to(DEV), mask.to(DEV))
     assert rel_err(s_gpu[:, :50], s_cpu[:, :50].to(DEV)) < 0.08
+
+
+# ---------------------------------------------------------------------------
+# column sum + fused linear (packed-QKV bias-grad path)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("shape", [(128, 1024), (1000, 3072)])
+def test_col_sum(dtype, shape):
+    torch.manual_seed(0)
+    x = torch.randn(shape, device=DEV, dtype=dtype)
+    out = ext().col_sum(x)
+    ref_out = x.float().sum(dim=0)
+    assert rel_err(out, ref_out) < (1e-5 if dtype == torch.float32 else 2e-2)
+
+
+def test_fused_linear_matches_flinear():
+    from bert_pytorch_amd.ops import fused_linear
+
+    torch.manual_seed(0)
+    x = torch.randn(64, 8, 256, device=DEV, requires_grad=True)
+    w = torch.randn(768, 256, device=DEV, requires_grad=True)
+    b = torch.randn(768, device=DEV, requires_grad=True)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y = fused_linear(x, w, b)
+        y.float().pow(2).mean().backward()
+    gx, gw, gb = x.grad.clone(), w.grad.clone(), b.grad.clone()
+    x.grad = w.grad = b.grad = None
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y2 = torch.nn.functional.linear(x, w, b)
+        y2.float().pow(2).mean().backward()
+    assert rel_err(y, y2) < 1e-6
+    assert rel_err(gx, x.grad) < 2e-2
+    assert rel_err(gw, w.grad) < 2e-2
+    assert rel_err(gb, b.grad) < 2e-2
+
+
+def test_fused_linear_pure_bf16():
+    from bert_pytorch_amd.ops import fused_linear
+
+    torch.manual_seed(0)
+    x = torch.randn(32, 128, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(64, 128, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    b = torch.randn(64, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    y = fused_linear(x, w, b)
+    y.float().sum().backward()
+    assert b.grad.dtype == torch.bfloat16
+    assert rel_err(b.grad, torch.full_like(b, 32.0)) < 2e-2
