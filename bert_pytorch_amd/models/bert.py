@@ -57,12 +57,18 @@ class LinearActivation(nn.Module):
         nn.init.uniform_(self.bias, -bound, bound)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if self.act == "gelu":
+        # activation names per the reference ACT2FN (src/modeling.py:139):
+        # gelu / bias_gelu are the fused HIP path; tanh/relu/swish eager.
+        if self.act in ("gelu", "bias_gelu"):
             y = F.linear(x, self.weight, None)
             return ops.fused_bias_gelu(y, self.bias)
         y = F.linear(x, self.weight, self.bias)
-        if self.act == "tanh":
+        if self.act in ("tanh", "bias_tanh"):
             return torch.tanh(y)
+        if self.act == "relu":
+            return F.relu(y)
+        if self.act == "swish":
+            return y * torch.sigmoid(y)
         raise ValueError(f"unsupported activation {self.act!r}")
 
 
