@@ -20,9 +20,7 @@
 
 namespace bpa {
 
-// declared in layernorm.hip
-__global__ void col_reduce_kernel(const float* __restrict__ parts, int nparts,
-                                  int H, float* __restrict__ out);
+torch::Tensor col_reduce_full(torch::Tensor parts);  // layernorm.hip
 
 template <typename T, int VEC>
 __global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
@@ -191,7 +189,7 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
   const int rows_per_block = 16;
   const int stripe_count =
       static_cast<int>((rows + rows_per_block - 1) / rows_per_block);
-  auto dbias = torch::empty({H}, fopts);
+  torch::Tensor dbias;
   auto dy_c = dy.contiguous();
   DISPATCH_FLOATING2(x.scalar_type(), "bias_gelu_bwd", [&] {
     TORCH_CHECK(H % kVec == 0, "bias_gelu_bwd: H % ", kVec, " != 0");
@@ -207,10 +205,7 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
                        reinterpret_cast<scalar_t*>(dx.data_ptr()),
                        part.data_ptr<float>(), static_cast<int>(rows), H,
                        rows_per_block);
-    dim3 rgrid((H + 255) / 256);
-    hipLaunchKernelGGL(col_reduce_kernel, rgrid, dim3(256), 0, stream,
-                       part.data_ptr<float>(), stripe_count, H,
-                       dbias.data_ptr<float>());
+    dbias = col_reduce_full(part);
   });
   if (bias.scalar_type() != torch::kFloat32) {
     return {dx, dbias.to(bias.scalar_type())};
@@ -225,7 +220,7 @@ torch::Tensor col_sum(torch::Tensor x) {
   const int rows_per_chunk = 32;
   const int n_chunks = (rows + rows_per_chunk - 1) / rows_per_chunk;
   auto fopts = x.options().dtype(torch::kFloat32);
-  auto out = torch::empty({H}, fopts);
+  torch::Tensor out;
   auto stream = at::hip::getCurrentHIPStream();
   DISPATCH_FLOATING2(x.scalar_type(), "col_sum", [&] {
     TORCH_CHECK(H % kVec == 0, "col_sum: H % ", kVec, " != 0");
@@ -237,9 +232,7 @@ torch::Tensor col_sum(torch::Tensor x) {
                        dim3(threads), 0, stream,
                        reinterpret_cast<const scalar_t*>(x.data_ptr()),
                        part.data_ptr<float>(), rows, H, rows_per_chunk);
-    hipLaunchKernelGGL(col_reduce_kernel, dim3((H + 255) / 256), dim3(256),
-                       0, stream, part.data_ptr<float>(), n_chunks, H,
-                       out.data_ptr<float>());
+    out = col_reduce_full(part);
   });
   return out;
 }

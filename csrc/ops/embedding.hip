@@ -16,8 +16,7 @@
 
 namespace bpa {
 
-__global__ void col_reduce_kernel(const float* __restrict__ parts, int nparts,
-                                  int H, float* __restrict__ out);
+torch::Tensor col_reduce_full(torch::Tensor parts);  // layernorm.hip
 
 template <typename T, typename TW, int VEC, bool HAS_TOK, bool TRAIN_DROP>
 __global__ void embed_fwd_kernel(
@@ -385,16 +384,8 @@ std::vector<torch::Tensor> embedding_ln_dropout_bwd(
   auto d_pos = torch::empty({max_pos, H}, fopts);
   auto d_tok = has_tok ? torch::empty({n_types, H}, fopts)
                        : torch::empty({0}, fopts);
-  auto dgamma = torch::empty({H}, fopts);
-  auto dbeta = torch::empty({H}, fopts);
-
-  dim3 rgrid((H + 255) / 256), rblock(256);
-  hipLaunchKernelGGL(col_reduce_kernel, rgrid, rblock, 0, stream,
-                     part_g.data_ptr<float>(), nblocks, H,
-                     dgamma.data_ptr<float>());
-  hipLaunchKernelGGL(col_reduce_kernel, rgrid, rblock, 0, stream,
-                     part_b.data_ptr<float>(), nblocks, H,
-                     dbeta.data_ptr<float>());
+  auto dgamma = col_reduce_full(part_g);
+  auto dbeta = col_reduce_full(part_b);
 
   const int64_t total = static_cast<int64_t>(rows) * H;
   const int sblocks = static_cast<int>(std::min<int64_t>((total + 255) / 256, 2048));
@@ -415,12 +406,7 @@ std::vector<torch::Tensor> embedding_ln_dropout_bwd(
                        dz.data_ptr<float>(), tt_c.data_ptr<int64_t>(),
                        tparts.data_ptr<float>(), rows, H,
                        static_cast<int>(n_types), rows_per_block);
-    dim3 tgrid((n_types * H + 255) / 256);
-    hipLaunchKernelGGL(col_reduce_kernel, tgrid, rblock, 0, stream,
-                       tparts.data_ptr<float>(), nblocks,
-                       static_cast<int>(n_types * H),
-                       d_tok.data_ptr<float>());
-    d_tok = d_tok.view({n_types, H});
+    d_tok = col_reduce_full(tparts).view({n_types, H});
   }
   return {d_word, d_pos, d_tok, dgamma, dbeta};
 }

@@ -22,8 +22,7 @@
 
 namespace bpa {
 
-__global__ void col_reduce_kernel(const float* __restrict__ parts, int nparts,
-                                  int H, float* __restrict__ out);
+torch::Tensor col_reduce_full(torch::Tensor parts);  // layernorm.hip
 
 template <typename T, int VEC, int ITEMS, bool HAS_BIAS, bool TRAIN_DROP>
 __global__ void bdrl_fwd_kernel(
@@ -290,7 +289,6 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
   const int n_slabs = has_bias ? 3 : 2;
   auto fopts = z.options().dtype(torch::kFloat32);
   auto part = torch::empty({n_chunks, n_slabs * H}, fopts);
-  auto out = torch::empty({n_slabs * H}, fopts);
   const bool train_drop = p > 0.0;
   auto stream = at::hip::getCurrentHIPStream();
   DISPATCH_T(z.scalar_type(), "bdrl_bwd", [&] {
@@ -337,10 +335,7 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
         part.data_ptr<float>(), rows, H, rows_per_chunk, n_chunks, has_bias,
         stream);
   });
-  dim3 rgrid((n_slabs * H + 255) / 256);
-  hipLaunchKernelGGL(col_reduce_kernel, rgrid, dim3(256), 0, stream,
-                     part.data_ptr<float>(), n_chunks, n_slabs * H,
-                     out.data_ptr<float>());
+  auto out = col_reduce_full(part);
   auto dgamma = out.narrow(0, 0, H).contiguous();
   auto dbeta = out.narrow(0, H, H).contiguous();
   auto dbias =

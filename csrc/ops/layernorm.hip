@@ -207,23 +207,45 @@ template void launch_col_stats<float, 4>(const float*, const float*,
                                          const float*, float*, int, int, int,
                                          int, bool, hipStream_t);
 
-// Deterministic single-pass column reduce: grid (ceil(H/256)); each
-// thread owns one output column and streams all `nparts` partial rows
-// (coalesced across lanes, loads pipeline across iterations). Direct
-// store - no atomics, so `out` needs NO zero-fill (the old atomic
-// version forced a zeros() kernel launch per call site).
-__global__ void col_reduce_kernel(const float* __restrict__ parts, int nparts,
-                                  int H, float* __restrict__ out) {
+// Deterministic strided column reduce: out[j][c] = sum over rows
+// {j, j+J, j+2J, ...} of parts[p][c]. Direct store, no atomics (so the
+// destination needs no zero-fill kernel), coalesced across lanes.
+// col_reduce_full runs it twice (J=16 then J=1) so both passes have
+// enough blocks in flight; a single direct pass had only ~S/256 blocks
+// and was 4x slower than the work it was reducing.
+__global__ void col_reduce_strided_kernel(const float* __restrict__ parts,
+                                          int nparts, int S, int J,
+                                          float* __restrict__ out) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= H) return;
+  if (c >= S) return;
+  const int j = blockIdx.y;
   float acc = 0.f;
-  int p = 0;
-  for (; p + 4 <= nparts; p += 4) {
-    const int64_t b = static_cast<int64_t>(p) * H + c;
-    acc += parts[b] + parts[b + H] + parts[b + 2 * H] + parts[b + 3 * H];
+  for (int p = j; p < nparts; p += J)
+    acc += parts[static_cast<int64_t>(p) * S + c];
+  out[static_cast<int64_t>(j) * S + c] = acc;
+}
+
+torch::Tensor col_reduce_full(torch::Tensor parts) {
+  const int nparts = parts.size(0);
+  const int S = parts.size(1);
+  auto out = torch::empty({S}, parts.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  const int xblocks = (S + 255) / 256;
+  if (nparts <= 32) {
+    hipLaunchKernelGGL(col_reduce_strided_kernel, dim3(xblocks, 1),
+                       dim3(256), 0, stream, parts.data_ptr<float>(), nparts,
+                       S, 1, out.data_ptr<float>());
+    return out;
   }
-  for (; p < nparts; ++p) acc += parts[static_cast<int64_t>(p) * H + c];
-  out[c] = acc;
+  constexpr int J = 16;
+  auto tmp = torch::empty({J, S}, parts.options());
+  hipLaunchKernelGGL(col_reduce_strided_kernel, dim3(xblocks, J), dim3(256),
+                     0, stream, parts.data_ptr<float>(), nparts, S, J,
+                     tmp.data_ptr<float>());
+  hipLaunchKernelGGL(col_reduce_strided_kernel, dim3(xblocks, 1), dim3(256),
+                     0, stream, tmp.data_ptr<float>(), J, S, 1,
+                     out.data_ptr<float>());
+  return out;
 }
 
 #define DISPATCH_FLOATING(TYPE, NAME, ...)                                   \
@@ -299,7 +321,6 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x,
   const int n_chunks = (rows + rows_per_chunk - 1) / rows_per_chunk;
   auto opts = x.options().dtype(torch::kFloat32);
   auto part = torch::empty({n_chunks, 2 * H}, opts);
-  auto dgb = torch::empty({2 * H}, opts);
   auto stream = at::hip::getCurrentHIPStream();
   auto dy_c = dy.contiguous();
   DISPATCH_FLOATING(x.scalar_type(), "ln_bwd", [&] {
@@ -321,10 +342,7 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x,
         part.data_ptr<float>(), rows, H, rows_per_chunk, n_chunks, false,
         stream);
   });
-  dim3 rgrid((2 * H + 255) / 256);
-  hipLaunchKernelGGL(col_reduce_kernel, rgrid, dim3(256), 0, stream,
-                     part.data_ptr<float>(), n_chunks, 2 * H,
-                     dgb.data_ptr<float>());
+  auto dgb = col_reduce_full(part);
   auto dgamma = dgb.narrow(0, 0, H);
   auto dbeta = dgb.narrow(0, H, H);
   if (gamma.scalar_type() != torch::kFloat32) {
