@@ -207,22 +207,33 @@ template void launch_col_stats<float, 4>(const float*, const float*,
                                          const float*, float*, int, int, int,
                                          int, bool, hipStream_t);
 
-// Deterministic strided column reduce: out[j][c] = sum over rows
-// {j, j+J, j+2J, ...} of parts[p][c]. Direct store, no atomics (so the
-// destination needs no zero-fill kernel), coalesced across lanes.
-// col_reduce_full runs it twice (J=16 then J=1) so both passes have
-// enough blocks in flight; a single direct pass had only ~S/256 blocks
-// and was 4x slower than the work it was reducing.
-__global__ void col_reduce_strided_kernel(const float* __restrict__ parts,
-                                          int nparts, int S, int J,
+// Deterministic blocked column reduce: out[j][c] = sum over the j-th
+// contiguous row block of parts[p][c]. Each thread owns 4 consecutive
+// fp32 columns (one uint4 per row) and streams its rows sequentially —
+// coalesced across lanes AND sequential in p, so the sweep runs at HBM
+// rate (the first strided-rows version thrashed; a single direct pass
+// had only ~S/256 blocks). Direct store: no atomics, no zero-fill.
+__global__ void col_reduce_blocked_kernel(const float* __restrict__ parts,
+                                          int nparts, int S, int rows_per_j,
                                           float* __restrict__ out) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  const int c = (blockIdx.x * blockDim.x + threadIdx.x) * 4;
   if (c >= S) return;
-  const int j = blockIdx.y;
-  float acc = 0.f;
-  for (int p = j; p < nparts; p += J)
-    acc += parts[static_cast<int64_t>(p) * S + c];
-  out[static_cast<int64_t>(j) * S + c] = acc;
+  const int p0 = blockIdx.y * rows_per_j;
+  const int p1 = min(p0 + rows_per_j, nparts);
+  float acc[4] = {};
+  for (int p = p0; p < p1; ++p) {
+    const float* src = parts + static_cast<int64_t>(p) * S + c;
+    if (c + 4 <= S) {
+      float v[4];
+      *reinterpret_cast<uint4*>(v) = *reinterpret_cast<const uint4*>(src);
+#pragma unroll
+      for (int k = 0; k < 4; ++k) acc[k] += v[k];
+    } else {
+      for (int k = 0; c + k < S; ++k) acc[k] += src[k];
+    }
+  }
+  float* dst = out + static_cast<int64_t>(blockIdx.y) * S + c;
+  for (int k = 0; k < 4 && c + k < S; ++k) dst[k] = acc[k];
 }
 
 torch::Tensor col_reduce_full(torch::Tensor parts) {
@@ -230,20 +241,21 @@ torch::Tensor col_reduce_full(torch::Tensor parts) {
   const int S = parts.size(1);
   auto out = torch::empty({S}, parts.options());
   auto stream = at::hip::getCurrentHIPStream();
-  const int xblocks = (S + 255) / 256;
-  if (nparts <= 32) {
-    hipLaunchKernelGGL(col_reduce_strided_kernel, dim3(xblocks, 1),
+  const int xblocks = (S / 4 + 255) / 256;
+  if (nparts <= 48) {
+    hipLaunchKernelGGL(col_reduce_blocked_kernel, dim3(xblocks, 1),
                        dim3(256), 0, stream, parts.data_ptr<float>(), nparts,
-                       S, 1, out.data_ptr<float>());
+                       S, nparts, out.data_ptr<float>());
     return out;
   }
-  constexpr int J = 16;
+  constexpr int J = 48;  // first-pass row groups
+  const int rows_per_j = (nparts + J - 1) / J;
   auto tmp = torch::empty({J, S}, parts.options());
-  hipLaunchKernelGGL(col_reduce_strided_kernel, dim3(xblocks, J), dim3(256),
-                     0, stream, parts.data_ptr<float>(), nparts, S, J,
-                     tmp.data_ptr<float>());
-  hipLaunchKernelGGL(col_reduce_strided_kernel, dim3(xblocks, 1), dim3(256),
-                     0, stream, tmp.data_ptr<float>(), J, S, 1,
+  hipLaunchKernelGGL(col_reduce_blocked_kernel, dim3(xblocks, J), dim3(256),
+                     0, stream, parts.data_ptr<float>(), nparts, S,
+                     rows_per_j, tmp.data_ptr<float>());
+  hipLaunchKernelGGL(col_reduce_blocked_kernel, dim3(xblocks, 1), dim3(256),
+                     0, stream, tmp.data_ptr<float>(), J, S, J,
                      out.data_ptr<float>());
   return out;
 }
