@@ -180,7 +180,9 @@ def main():
             "dtype": "bf16" if use_cuda else "fp32(cpu)",
             "data": "synthetic",
             "config": {
-                "model": "bert-large-uncased (24L/1024H/16h, vocab 30528)",
+                "model": f"{os.path.splitext(os.path.basename(args.model_config))[0]} "
+                         f"({config.num_hidden_layers}L/{config.hidden_size}H/"
+                         f"{config.num_attention_heads}h, vocab {config.vocab_size})",
                 "phase": args.phase,
                 "global_batch": world * bsz * accum,
                 "named_global_batch": phase["named_global"],
