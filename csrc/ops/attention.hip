@@ -66,41 +66,6 @@ __device__ __forceinline__ float uniform_at(const Philox& ph, uint64_t ctr,
   return u32_to_uniform(r4[word]);
 }
 
-// Stage a 64x64 bf16 tile TRANSPOSED into LDS (dst[col][row], row pitch
-// kStride) from global row-major rows (pitch src_stride elements).
-// 256 threads, one 4x4 block each: 4 coalesced 8-B global loads
-// (16-lane groups cover 128 contiguous bytes), register transpose,
-// 4 8-B ds_writes. PMC showed the previous per-element b16 column
-// writes at 4-8-way bank conflict (~53% of LDS cycles); b64 writes
-// land at worst 2-way.
-__device__ __forceinline__ void stage_transposed_64x64(
-    const __bf16* __restrict__ src, int64_t src_stride, int valid_rows,
-    __bf16* __restrict__ dst) {
-  const int t = threadIdx.x;
-  const int br = t >> 4;      // 4-row block index (0..15)
-  const int bc = t & 15;      // 4-col block index (0..15)
-  __bf16 blk[4][4];
-#pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    const int r = br * 4 + j;
-    if (r < valid_rows) {
-      *reinterpret_cast<uint2*>(blk[j]) = *reinterpret_cast<const uint2*>(
-          src + static_cast<int64_t>(r) * src_stride + bc * 4);
-    } else {
-#pragma unroll
-      for (int k = 0; k < 4; ++k) blk[j][k] = __bf16(0.f);
-    }
-  }
-#pragma unroll
-  for (int jj = 0; jj < 4; ++jj) {
-    __bf16 col[4];
-#pragma unroll
-    for (int j = 0; j < 4; ++j) col[j] = blk[j][jj];
-    *reinterpret_cast<uint2*>(&dst[(bc * 4 + jj) * kStride + br * 4]) =
-        *reinterpret_cast<const uint2*>(col);
-  }
-}
-
 // ---------------------------------------------------------------------------
 // forward
 // ---------------------------------------------------------------------------
@@ -149,7 +114,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   for (int kt = 0; kt < n_kv; ++kt) {
     const int k0 = kt * 64;
     __syncthreads();
-    // stage K natural (4 threads/row, uint4) + V transposed (4x4 blocks)
+    // stage K natural + V transposed; 4 threads per row, 16 cols each
     {
       const int row = tid >> 2, colc = (tid & 3) * 16;
       const int krow = k0 + row;
@@ -158,13 +123,21 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
             reinterpret_cast<const uint4*>(kbase + static_cast<int64_t>(krow) * rs3 + colc);
         *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc]) = src[0];
         *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc + 8]) = src[1];
+        __bf16 vv[16];
+        const uint4* vsrc =
+            reinterpret_cast<const uint4*>(vbase + static_cast<int64_t>(krow) * rs3 + colc);
+        *reinterpret_cast<uint4*>(&vv[0]) = vsrc[0];
+        *reinterpret_cast<uint4*>(&vv[8]) = vsrc[1];
+#pragma unroll
+        for (int j = 0; j < 16; ++j) Vt_lds[(colc + j) * kStride + row] = vv[j];
       } else {
         uint4 zero{0, 0, 0, 0};
         *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc]) = zero;
         *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc + 8]) = zero;
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          Vt_lds[(colc + j) * kStride + row] = __bf16(0.f);
       }
-      stage_transposed_64x64(vbase + static_cast<int64_t>(k0) * rs3, rs3,
-                             S - k0, Vt_lds);
     }
     __syncthreads();
 
@@ -377,24 +350,29 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
     {
       const int row = tid >> 2, colc = (tid & 3) * 16;
       const int krow = k0 + row;
+      __bf16 kv[16];
       if (krow < S) {
         const uint4* ks = reinterpret_cast<const uint4*>(
             kbase + static_cast<int64_t>(krow) * rs3 + colc);
-        *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc]) = ks[0];
-        *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc + 8]) = ks[1];
+        *reinterpret_cast<uint4*>(&kv[0]) = ks[0];
+        *reinterpret_cast<uint4*>(&kv[8]) = ks[1];
         const uint4* vs = reinterpret_cast<const uint4*>(
             vbase + static_cast<int64_t>(krow) * rs3 + colc);
         *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc]) = vs[0];
         *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc + 8]) = vs[1];
       } else {
         uint4 zero{0, 0, 0, 0};
-        *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc]) = zero;
-        *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc + 8]) = zero;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) kv[j] = __bf16(0.f);
         *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc]) = zero;
         *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc + 8]) = zero;
       }
-      stage_transposed_64x64(kbase + static_cast<int64_t>(k0) * rs3, rs3,
-                             S - k0, Kt_lds);
+      *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc]) =
+          *reinterpret_cast<uint4*>(&kv[0]);
+      *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc + 8]) =
+          *reinterpret_cast<uint4*>(&kv[8]);
+#pragma unroll
+      for (int j = 0; j < 16; ++j) Kt_lds[(colc + j) * kStride + row] = kv[j];
     }
     __syncthreads();
 
@@ -537,30 +515,37 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   for (int qt = 0; qt < n_q; ++qt) {
     const int q0 = qt * 64;
     __syncthreads();
-    // stage Q, dO natural (uint4) + Q^T, dO^T (4x4 block transpose)
+    // stage Q, Q^T, dO, dO^T + lse/delta for this q-tile
     {
       const int row = tid >> 2, colc = (tid & 3) * 16;
       const int qrow = q0 + row;
+      __bf16 qv[16], dov[16];
       if (qrow < S) {
         const uint4* qs =
             reinterpret_cast<const uint4*>(qbase + static_cast<int64_t>(qrow) * rs3 + colc);
-        *reinterpret_cast<uint4*>(&Q_lds[row * kStride + colc]) = qs[0];
-        *reinterpret_cast<uint4*>(&Q_lds[row * kStride + colc + 8]) = qs[1];
+        *reinterpret_cast<uint4*>(&qv[0]) = qs[0];
+        *reinterpret_cast<uint4*>(&qv[8]) = qs[1];
         const uint4* ds =
             reinterpret_cast<const uint4*>(dobase + static_cast<int64_t>(qrow) * H + colc);
-        *reinterpret_cast<uint4*>(&dO_lds[row * kStride + colc]) = ds[0];
-        *reinterpret_cast<uint4*>(&dO_lds[row * kStride + colc + 8]) = ds[1];
+        *reinterpret_cast<uint4*>(&dov[0]) = ds[0];
+        *reinterpret_cast<uint4*>(&dov[8]) = ds[1];
       } else {
-        uint4 zero{0, 0, 0, 0};
-        *reinterpret_cast<uint4*>(&Q_lds[row * kStride + colc]) = zero;
-        *reinterpret_cast<uint4*>(&Q_lds[row * kStride + colc + 8]) = zero;
-        *reinterpret_cast<uint4*>(&dO_lds[row * kStride + colc]) = zero;
-        *reinterpret_cast<uint4*>(&dO_lds[row * kStride + colc + 8]) = zero;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) qv[j] = dov[j] = __bf16(0.f);
       }
-      stage_transposed_64x64(qbase + static_cast<int64_t>(q0) * rs3, rs3,
-                             S - q0, Qt_lds);
-      stage_transposed_64x64(dobase + static_cast<int64_t>(q0) * H, H,
-                             S - q0, dOt_lds);
+      *reinterpret_cast<uint4*>(&Q_lds[row * kStride + colc]) =
+          *reinterpret_cast<uint4*>(&qv[0]);
+      *reinterpret_cast<uint4*>(&Q_lds[row * kStride + colc + 8]) =
+          *reinterpret_cast<uint4*>(&qv[8]);
+      *reinterpret_cast<uint4*>(&dO_lds[row * kStride + colc]) =
+          *reinterpret_cast<uint4*>(&dov[0]);
+      *reinterpret_cast<uint4*>(&dO_lds[row * kStride + colc + 8]) =
+          *reinterpret_cast<uint4*>(&dov[8]);
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        Qt_lds[(colc + j) * kStride + row] = qv[j];
+        dOt_lds[(colc + j) * kStride + row] = dov[j];
+      }
       if (tid < 64) {
         const int qr = q0 + tid;
         lse_lds[tid] = (qr < S) ? lse[static_cast<int64_t>(bh) * S + qr] : 0.f;
