@@ -5,7 +5,7 @@
 # dataloader H2D, fused kernels, FusedLAMB, checkpoint IO and the
 # two-phase optimizer-state surgery on device.
 set -euo pipefail
-OUT="${1:-gpurun_out/e2e}"
+OUT="${1:-/tmp/bpa_e2e}"  # keep checkpoints OFF gpurun_out (64 MiB merge limit)
 rm -rf "$OUT"; mkdir -p "$OUT/data128" "$OUT/data512"
 
 python - <<EOF

@@ -43,6 +43,18 @@ class _FusedAttention(torch.autograd.Function):
         return dqkv, None, None, None, None
 
 
+def _kernel_supported(qkv: torch.Tensor, num_heads: int) -> bool:
+    """The gfx950 MFMA kernel covers the BERT/RoBERTa operating points:
+    bf16, head_dim 64, S % 16 == 0. Other shapes (tiny test configs,
+    fp32-without-autocast) run the eager composite on device."""
+    head_dim = qkv.shape[-1] // 3 // num_heads
+    return (
+        qkv.dtype == torch.bfloat16
+        and head_dim == 64
+        and qkv.shape[1] % 16 == 0
+    )
+
+
 def fused_attention(
     qkv: torch.Tensor,
     seqlens: torch.Tensor,
@@ -51,6 +63,6 @@ def fused_attention(
     training: bool,
 ) -> torch.Tensor:
     """qkv [B,S,3H] packed, seqlens [B] int32 -> context [B,S,H]."""
-    if use_native(qkv):
+    if use_native(qkv) and _kernel_supported(qkv, num_heads):
         return _FusedAttention.apply(qkv, seqlens, num_heads, p, training)
     return _reference.attention(qkv, seqlens, num_heads, p, training)
