@@ -20,6 +20,7 @@ Launch: python bench.py --gpus N --steps K --warmup W
 from __future__ import annotations
 
 import argparse
+import contextlib
 import json
 import os
 import sys
@@ -58,6 +59,9 @@ def parse_args():
     p.add_argument("--local_batch", type=int, default=0)
     p.add_argument("--model_config", type=str,
                    default="config/bert_large_uncased_config.json")
+    p.add_argument("--window_autocast", action="store_true",
+                   help="hold one autocast region per accumulation window "
+                        "so weight-cast caching spans its micro-steps")
     p.add_argument("--pure_bf16", action="store_true",
                    help="bf16 model weights + fp32 LAMB masters instead of "
                         "fp32 weights + autocast (no per-microbatch weight "
@@ -124,12 +128,41 @@ def main():
     gen = torch.Generator(device=device).manual_seed(args.seed + rank)
     vocab_unpadded = 30522
 
+    ac_enabled = use_cuda and not args.pure_bf16
+
+    class WindowAutocast:
+        """Hold one autocast region open across an accumulation window
+        so torch's weight-cast cache is reused for all its micro-steps
+        (weights only change at the optimizer step, i.e. the window
+        boundary). Saves the per-microbatch fp32->bf16 weight casts."""
+
+        def __init__(self):
+            self._ctx = None
+
+        def cycle(self, micro_idx: int) -> None:
+            if not (ac_enabled and args.window_autocast):
+                return
+            if micro_idx % accum == 0:
+                self.close()
+                self._ctx = torch.autocast(device.type, dtype=torch.bfloat16)
+                self._ctx.__enter__()
+
+        def close(self) -> None:
+            if self._ctx is not None:
+                self._ctx.__exit__(None, None, None)
+                self._ctx = None
+
+    wa = WindowAutocast()
+    per_step_ac = ac_enabled and not args.window_autocast
+
     def one_step(micro_idx: int) -> None:
         batch = make_batch(gen, device, bsz, seq, vocab_unpadded, phase["max_pred"])
         ids, tt, mask, labels, nsp = batch
         sync = (micro_idx + 1) % accum == 0
+        wa.cycle(micro_idx)
         with torch.autocast(device.type, dtype=torch.bfloat16,
-                            enabled=use_cuda and not args.pure_bf16):
+                            enabled=per_step_ac) if per_step_ac or not ac_enabled \
+                else contextlib.nullcontext():
             scores, rel, glabels = model(ids, tt, mask, masked_lm_labels=labels)
             loss = criterion(scores, rel, glabels, nsp) / accum
         if sync or not isinstance(model, torch.nn.parallel.DistributedDataParallel):
@@ -145,6 +178,7 @@ def main():
     model.train()
     for i in range(args.warmup):
         one_step(i)
+    wa.close()
 
     comm.barrier()
     if use_cuda:
@@ -154,6 +188,7 @@ def main():
         one_step(i)
     if use_cuda:
         torch.cuda.synchronize()
+    wa.close()
     comm.barrier()
     elapsed = time.perf_counter() - t0
 

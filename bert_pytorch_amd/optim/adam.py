@@ -50,8 +50,12 @@ class FusedAdam(Optimizer):
 
     def state_dict(self):  # noqa: D102
         sd = super().state_dict()
-        for s in sd["state"].values():  # grad32 is per-step scratch
-            s.pop("grad32", None)
+        # drop per-step scratch WITHOUT mutating the live state (torch
+        # packs references to the optimizer's state dicts, not copies)
+        sd["state"] = {
+            k: {kk: vv for kk, vv in v.items() if kk != "grad32"}
+            for k, v in sd["state"].items()
+        }
         return sd
 
     def load_state_dict(self, state_dict):  # noqa: D102
