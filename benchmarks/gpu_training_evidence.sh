@@ -14,7 +14,7 @@ from bert_pytorch_amd.data import synth
 synth.make_dataset("$WORK/d128", num_shards=2, samples_per_shard=4096,
                    seq_len=128, vocab_size=30522, seed=0)
 synth.make_dataset("$WORK/d128nonsp", num_shards=1, samples_per_shard=2048,
-                   seq_len=128, vocab_size=30522, seed=1, nsp=False)
+                   seq_len=128, vocab_size=28996, seed=1, nsp=False)
 EOF
 
 # 1) convergence: BERT-base, 300 steps
