@@ -284,7 +284,7 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
   auto dy_c = dy.contiguous();
   auto dx = torch::empty_like(z);
   auto dz_res = torch::empty_like(z);
-  const int rows_per_chunk = 8;
+  const int rows_per_chunk = 16;  // halves partial-buffer traffic; 16 independent loads/thread
   const int n_chunks = (rows + rows_per_chunk - 1) / rows_per_chunk;
   const int n_slabs = has_bias ? 3 : 2;
   auto fopts = z.options().dtype(torch::kFloat32);

@@ -352,7 +352,7 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x,
   const int rows = x.size(0), H = x.size(1);
   auto gamma_f = gamma.contiguous().to(torch::kFloat32);
   auto dx = torch::empty_like(x);
-  const int rows_per_chunk = 8;
+  const int rows_per_chunk = 16;  // halves partial-buffer traffic; 16 independent loads/thread
   const int n_chunks = (rows + rows_per_chunk - 1) / rows_per_chunk;
   auto opts = x.options().dtype(torch::kFloat32);
   auto part = torch::empty({n_chunks, 2 * H}, opts);
