@@ -74,9 +74,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const __bf16* __restrict__ qkv, const int* __restrict__ seqlens,
     __bf16* __restrict__ out, float* __restrict__ lse_out, int B, int S,
     int NH, float p, float scale, uint64_t seed, uint64_t offset) {
+  // 4 waves x 32 q-rows (two 16-row subtiles per wave): the K/V tile is
+  // staged once per 128 q-rows, each K/V fragment read feeds TWO
+  // independent MFMA chains, and the softmax work of the two subtiles
+  // interleaves - targeting the measured issue-stall/parked time.
   const int bh = blockIdx.y;
   const int b = bh / NH, h = bh % NH;
-  const int q0 = blockIdx.x * 64;
+  const int q0 = blockIdx.x * 128;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
   const int g = (lane >> 4), li = lane & 15;
@@ -95,20 +99,24 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* K_lds = reinterpret_cast<__bf16*>(smem);             // [64][72]
   __bf16* Vt_lds = K_lds + 64 * kStride;                       // [64][72]
-  float* alpha_lds = reinterpret_cast<float*>(Vt_lds + 64 * kStride);  // [4][16]
-  float* stat_lds = alpha_lds + 4 * 16;                        // [4][16]
+  float* alpha_lds = reinterpret_cast<float*>(Vt_lds + 64 * kStride);  // [4][2][16]
+  float* stat_lds = alpha_lds + 4 * 2 * 16;                    // [4][2][16]
 
-  // Q fragments for this wave's 16 rows (row = q0 + wave*16 + li), held
-  // in registers for the whole K/V sweep.
-  const int q_row = q0 + wave * 16 + li;
-  const int q_ld = min(q_row, S - 1);
-  bf16x8 qfrag[2];
+  // this wave's two q-subtiles: rows q0 + wave*32 + sub*16 + li
+  int q_row[2];
+  bf16x8 qfrag[2][2];
 #pragma unroll
-  for (int c = 0; c < 2; ++c)
-    qfrag[c] = frag_row(qbase + static_cast<int64_t>(q_ld) * rs3, 32 * c, g);
+  for (int sub = 0; sub < 2; ++sub) {
+    q_row[sub] = q0 + wave * 32 + sub * 16 + li;
+    const int q_ld = min(q_row[sub], S - 1);
+#pragma unroll
+    for (int c = 0; c < 2; ++c)
+      qfrag[sub][c] =
+          frag_row(qbase + static_cast<int64_t>(q_ld) * rs3, 32 * c, g);
+  }
 
-  float m_run = -1e30f, l_run = 0.f;
-  f32x4 acc_o[4] = {};
+  float m_run[2] = {-1e30f, -1e30f}, l_run[2] = {0.f, 0.f};
+  f32x4 acc_o[2][4] = {};
 
   const int n_kv = (S + 63) / 64;
   for (int kt = 0; kt < n_kv; ++kt) {
@@ -141,118 +149,140 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     }
     __syncthreads();
 
-    // S^T tiles: A = K (16 keys), B = Q^T; C[key_local][q]: q = li
-    f32x4 s_acc[4];
+    // S^T tiles: one K-fragment pair feeds both subtiles' MFMA chains
+    f32x4 s_acc[2][4];
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
       const __bf16* krow = &K_lds[(t * 16 + li) * kStride];
-      f32x4 acc = {};
-      acc = MFMA16(frag_row(krow, 0, g), qfrag[0], acc);
-      acc = MFMA16(frag_row(krow, 32, g), qfrag[1], acc);
-      s_acc[t] = acc;
+      const bf16x8 kf0 = frag_row(krow, 0, g);
+      const bf16x8 kf1 = frag_row(krow, 32, g);
+      f32x4 a0 = {}, a1 = {};
+      a0 = MFMA16(kf0, qfrag[0][0], a0);
+      a1 = MFMA16(kf0, qfrag[1][0], a1);
+      a0 = MFMA16(kf1, qfrag[0][1], a0);
+      a1 = MFMA16(kf1, qfrag[1][1], a1);
+      s_acc[0][t] = a0;
+      s_acc[1][t] = a1;
     }
 
-    // mask + scale; per-q tile max (this lane holds keys k0+16t+g*4+r)
-    float sv[4][4];
-    float tmax = -1e30f;
+    float sv[2][4][4];
 #pragma unroll
-    for (int t = 0; t < 4; ++t) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int key = k0 + t * 16 + g * 4 + r;
-        const bool valid = key < slen && key < S;
-        sv[t][r] = valid ? s_acc[t][r] * scale : -1e30f;
-        tmax = fmaxf(tmax, sv[t][r]);
-      }
-    }
-    tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
-    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
-    const float m_new = fmaxf(m_run, tmax);
-    const float alpha = __expf(m_run - m_new);
-    float rowsum = 0.f;
-#pragma unroll
-    for (int t = 0; t < 4; ++t)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        sv[t][r] = __expf(sv[t][r] - m_new);
-        rowsum += sv[t][r];
-      }
-    rowsum += __shfl_xor(rowsum, 16, 64);
-    rowsum += __shfl_xor(rowsum, 32, 64);
-    l_run = l_run * alpha + rowsum;
-    m_run = m_new;
-
-    // O rescale: alpha indexed by this wave's q rows via LDS broadcast
-    if (g == 0) alpha_lds[wave * 16 + li] = alpha;
-    __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0): same-wave LDS visibility
-#pragma unroll
-    for (int n = 0; n < 4; ++n)
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        acc_o[n][r] *= alpha_lds[wave * 16 + g * 4 + r];
-
-    // dropout on the unnormalized probabilities (l keeps the full sum)
-    if (TRAIN_DROP) {
-      const int q_abs = q_row;
+    for (int sub = 0; sub < 2; ++sub) {
+      float tmax = -1e30f;
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
 #pragma unroll
-        for (int rpair = 0; rpair < 2; ++rpair) {
-          const int key = k0 + t * 16 + g * 4 + rpair * 2;
-          uint32_t r4[4];
-          philox(drop_base + static_cast<uint64_t>(q_abs >> 1) * s2 + (key >> 1),
-                 r4);
-          const int w0 = (q_abs & 1) * 2;
+        for (int r = 0; r < 4; ++r) {
+          const int key = k0 + t * 16 + g * 4 + r;
+          const bool valid = key < slen && key < S;
+          sv[sub][t][r] = valid ? s_acc[sub][t][r] * scale : -1e30f;
+          tmax = fmaxf(tmax, sv[sub][t][r]);
+        }
+      }
+      tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
+      tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+      const float m_new = fmaxf(m_run[sub], tmax);
+      const float alpha = __expf(m_run[sub] - m_new);
+      float rowsum = 0.f;
 #pragma unroll
-          for (int j = 0; j < 2; ++j) {
-            const bool keep = u32_to_uniform(r4[w0 + j]) >= p;
-            sv[t][rpair * 2 + j] =
-                keep ? sv[t][rpair * 2 + j] * inv_keep : 0.f;
+      for (int t = 0; t < 4; ++t)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          sv[sub][t][r] = __expf(sv[sub][t][r] - m_new);
+          rowsum += sv[sub][t][r];
+        }
+      rowsum += __shfl_xor(rowsum, 16, 64);
+      rowsum += __shfl_xor(rowsum, 32, 64);
+      l_run[sub] = l_run[sub] * alpha + rowsum;
+      m_run[sub] = m_new;
+      if (g == 0) alpha_lds[(wave * 2 + sub) * 16 + li] = alpha;
+    }
+    __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0): same-wave LDS visibility
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          acc_o[sub][n][r] *= alpha_lds[(wave * 2 + sub) * 16 + g * 4 + r];
+
+    if (TRAIN_DROP) {
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+        const int q_abs = q_row[sub];
+#pragma unroll
+        for (int t = 0; t < 4; ++t) {
+#pragma unroll
+          for (int rpair = 0; rpair < 2; ++rpair) {
+            const int key = k0 + t * 16 + g * 4 + rpair * 2;
+            uint32_t r4[4];
+            philox(drop_base +
+                       static_cast<uint64_t>(q_abs >> 1) * s2 + (key >> 1),
+                   r4);
+            const int w0 = (q_abs & 1) * 2;
+#pragma unroll
+            for (int j = 0; j < 2; ++j) {
+              const bool keep = u32_to_uniform(r4[w0 + j]) >= p;
+              sv[sub][t][rpair * 2 + j] =
+                  keep ? sv[sub][t][rpair * 2 + j] * inv_keep : 0.f;
+            }
           }
         }
       }
     }
 
-    // P fragments chain from sv: chunk c covers keys 32c..32c+31
-    bf16x8 pa[2];
+    // P fragments chain from sv; one V^T fragment feeds both subtiles
+    bf16x8 pa[2][2];
 #pragma unroll
-    for (int c = 0; c < 2; ++c) {
-      union {
-        bf16x8 v;
-        __bf16 e[8];
-      } pk;
+    for (int sub = 0; sub < 2; ++sub) {
 #pragma unroll
-      for (int e = 0; e < 8; ++e)
-        pk.e[e] = __bf16(sv[2 * c + (e >> 2)][e & 3]);
-      pa[c] = pk.v;
+      for (int c = 0; c < 2; ++c) {
+        union {
+          bf16x8 v;
+          __bf16 e[8];
+        } pk;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          pk.e[e] = __bf16(sv[sub][2 * c + (e >> 2)][e & 3]);
+        pa[sub][c] = pk.v;
+      }
     }
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
       const __bf16* vrow = &Vt_lds[(n * 16 + li) * kStride];
-      acc_o[n] = MFMA16(pa[0], frag_row(vrow, 0, g), acc_o[n]);
-      acc_o[n] = MFMA16(pa[1], frag_row(vrow, 32, g), acc_o[n]);
+      const bf16x8 vf0 = frag_row(vrow, 0, g);
+      const bf16x8 vf1 = frag_row(vrow, 32, g);
+      acc_o[0][n] = MFMA16(pa[0][0], vf0, acc_o[0][n]);
+      acc_o[1][n] = MFMA16(pa[1][0], vf0, acc_o[1][n]);
+      acc_o[0][n] = MFMA16(pa[0][1], vf1, acc_o[0][n]);
+      acc_o[1][n] = MFMA16(pa[1][1], vf1, acc_o[1][n]);
     }
   }
 
   // epilogue: normalize rows by l (broadcast per-wave through LDS)
-  if (g == 0) stat_lds[wave * 16 + li] = l_run;
+#pragma unroll
+  for (int sub = 0; sub < 2; ++sub)
+    if (g == 0) stat_lds[(wave * 2 + sub) * 16 + li] = l_run[sub];
   __builtin_amdgcn_s_waitcnt(0);
 #pragma unroll
-  for (int n = 0; n < 4; ++n) {
+  for (int sub = 0; sub < 2; ++sub) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int qr = q0 + wave * 16 + g * 4 + r;
-      if (qr < S) {
-        const float l = stat_lds[wave * 16 + g * 4 + r];
-        const float o = acc_o[n][r] / (l > 0.f ? l : 1.f);
-        out[(static_cast<int64_t>(b) * S + qr) * H + h * 64 + n * 16 + li] =
-            __bf16(o);
+    for (int n = 0; n < 4; ++n) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qr = q0 + wave * 32 + sub * 16 + g * 4 + r;
+        if (qr < S) {
+          const float l = stat_lds[(wave * 2 + sub) * 16 + g * 4 + r];
+          const float o = acc_o[sub][n][r] / (l > 0.f ? l : 1.f);
+          out[(static_cast<int64_t>(b) * S + qr) * H + h * 64 + n * 16 + li] =
+              __bf16(o);
+        }
       }
     }
+    if (g == 0 && q_row[sub] < S)
+      lse_out[static_cast<int64_t>(bh) * S + q_row[sub]] =
+          m_run[sub] + __logf(l_run[sub] > 0.f ? l_run[sub] : 1.f);
   }
-  if (g == 0 && wave * 16 + li + q0 < S && q_row < S)
-    lse_out[static_cast<int64_t>(bh) * S + q_row] =
-        m_run + __logf(l_run > 0.f ? l_run : 1.f);
 }
 
 // ---------------------------------------------------------------------------
@@ -667,8 +697,9 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor qkv,
   auto lse = torch::empty({B * NH, S}, qkv.options().dtype(torch::kFloat32));
   const float scale = 1.0f / sqrtf(64.f);
   auto stream = at::hip::getCurrentHIPStream();
-  dim3 grid((S + 63) / 64, B * NH), block(256);
-  const size_t lds = 2 * 64 * kStride * sizeof(__bf16) + 2 * 64 * sizeof(float);
+  dim3 grid((S + 127) / 128, B * NH), block(256);  // 128 q-rows per block
+  const size_t lds =
+      2 * 64 * kStride * sizeof(__bf16) + 4 * 64 * sizeof(float);
   const bool train_drop = p > 0.0;
   if (train_drop) {
     hipLaunchKernelGGL((attn_fwd_kernel<true>), grid, block, lds, stream,
