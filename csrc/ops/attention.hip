@@ -336,9 +336,12 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
     const int* __restrict__ seqlens, const float* __restrict__ lse,
     const float* __restrict__ delta, __bf16* __restrict__ dqkv, int B, int S,
     int NH, float p, float scale, uint64_t seed, uint64_t offset) {
+  // 4 waves x 32 q-rows (two 16-row subtiles per wave), mirroring
+  // attn_fwd_kernel: K/V/K^T staged once per 128 q-rows, every staged
+  // fragment feeds two independent MFMA chains.
   const int bh = blockIdx.y;
   const int b = bh / NH, h = bh % NH;
-  const int q0 = blockIdx.x * 64;
+  const int q0 = blockIdx.x * 128;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
   const int g = (lane >> 4), li = lane & 15;
@@ -359,19 +362,28 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
   __bf16* V_lds = K_lds + 64 * kStride;              // [64][72] natural
   __bf16* Kt_lds = V_lds + 64 * kStride;             // [64][72] transposed
 
-  // this wave's 16 q rows: Q and dO fragments + lse/delta, registers
-  const int q_row = q0 + wave * 16 + li;
-  const int q_ld = min(q_row, S - 1);
-  bf16x8 qfrag[2], dofrag[2];
+  // this wave's two q-subtiles: Q/dO fragments + lse/delta in registers
+  int q_row[2];
+  bf16x8 qfrag[2][2], dofrag[2][2];
+  float lse_q[2], dlt_q[2];
 #pragma unroll
-  for (int c = 0; c < 2; ++c) {
-    qfrag[c] = frag_row(qbase + static_cast<int64_t>(q_ld) * rs3, 32 * c, g);
-    dofrag[c] = frag_row(dobase + static_cast<int64_t>(q_ld) * H, 32 * c, g);
+  for (int sub = 0; sub < 2; ++sub) {
+    q_row[sub] = q0 + wave * 32 + sub * 16 + li;
+    const int q_ld = min(q_row[sub], S - 1);
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      qfrag[sub][c] =
+          frag_row(qbase + static_cast<int64_t>(q_ld) * rs3, 32 * c, g);
+      dofrag[sub][c] =
+          frag_row(dobase + static_cast<int64_t>(q_ld) * H, 32 * c, g);
+    }
+    lse_q[sub] =
+        (q_row[sub] < S) ? lse[static_cast<int64_t>(bh) * S + q_row[sub]] : 0.f;
+    dlt_q[sub] =
+        (q_row[sub] < S) ? delta[static_cast<int64_t>(bh) * S + q_row[sub]] : 0.f;
   }
-  const float lse_q = (q_row < S) ? lse[static_cast<int64_t>(bh) * S + q_row] : 0.f;
-  const float dlt_q = (q_row < S) ? delta[static_cast<int64_t>(bh) * S + q_row] : 0.f;
 
-  f32x4 acc_dq[4] = {};
+  f32x4 acc_dq[2][4] = {};
 
   const int n_kv = (S + 63) / 64;
   for (int kt = 0; kt < n_kv; ++kt) {
@@ -406,70 +418,89 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
     }
     __syncthreads();
 
-    // S^T and dP^T tiles (C[key][q], q = li): A = K rows / V rows
-    float dsv[4][4];
+    // S^T and dP^T tiles (C[key][q], q = li); K/V fragments shared
+    float dsv[2][4][4];
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
       const __bf16* krow = &K_lds[(t * 16 + li) * kStride];
       const __bf16* vrow = &V_lds[(t * 16 + li) * kStride];
-      f32x4 sacc = {}, dpacc = {};
-      sacc = MFMA16(frag_row(krow, 0, g), qfrag[0], sacc);
-      sacc = MFMA16(frag_row(krow, 32, g), qfrag[1], sacc);
-      dpacc = MFMA16(frag_row(vrow, 0, g), dofrag[0], dpacc);
-      dpacc = MFMA16(frag_row(vrow, 32, g), dofrag[1], dpacc);
-      // one Philox call serves a consecutive key pair (key>>1 shared)
-      uint32_t r4[4];
+      const bf16x8 kf0 = frag_row(krow, 0, g), kf1 = frag_row(krow, 32, g);
+      const bf16x8 vf0 = frag_row(vrow, 0, g), vf1 = frag_row(vrow, 32, g);
+      f32x4 sacc[2] = {}, dpacc[2] = {};
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int key = k0 + t * 16 + g * 4 + r;
-        const bool valid = key < slen && key < S && q_row < S;
-        const float pr =
-            valid ? __expf(sacc[r] * scale - lse_q) : 0.f;
-        float dpd = dpacc[r];
-        if (TRAIN_DROP) {
-          if ((r & 1) == 0) {
-            philox(drop_base + static_cast<uint64_t>(q_row >> 1) * s2 +
-                       (key >> 1),
-                   r4);
+      for (int sub = 0; sub < 2; ++sub) {
+        sacc[sub] = MFMA16(kf0, qfrag[sub][0], sacc[sub]);
+        sacc[sub] = MFMA16(kf1, qfrag[sub][1], sacc[sub]);
+        dpacc[sub] = MFMA16(vf0, dofrag[sub][0], dpacc[sub]);
+        dpacc[sub] = MFMA16(vf1, dofrag[sub][1], dpacc[sub]);
+      }
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+        // one Philox call serves a consecutive key pair (key>>1 shared)
+        uint32_t r4[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int key = k0 + t * 16 + g * 4 + r;
+          const bool valid = key < slen && key < S && q_row[sub] < S;
+          const float pr =
+              valid ? __expf(sacc[sub][r] * scale - lse_q[sub]) : 0.f;
+          float dpd = dpacc[sub][r];
+          if (TRAIN_DROP) {
+            if ((r & 1) == 0) {
+              philox(drop_base +
+                         static_cast<uint64_t>(q_row[sub] >> 1) * s2 +
+                         (key >> 1),
+                     r4);
+            }
+            const bool keep =
+                u32_to_uniform(r4[(q_row[sub] & 1) * 2 + (key & 1)]) >= p;
+            dpd = keep ? dpd * inv_keep : 0.f;
           }
-          const bool keep =
-              u32_to_uniform(r4[(q_row & 1) * 2 + (key & 1)]) >= p;
-          dpd = keep ? dpd * inv_keep : 0.f;
+          dsv[sub][t][r] = pr * (dpd - dlt_q[sub]) * scale;
         }
-        dsv[t][r] = pr * (dpd - dlt_q) * scale;
       }
     }
 
-    // chain dS^T into A-fragments over keys (chunk c: keys 32c..32c+31)
-    bf16x8 dsfrag[2];
+    // chain dS^T into A-fragments over keys; K^T fragments shared
+    bf16x8 dsfrag[2][2];
 #pragma unroll
-    for (int c = 0; c < 2; ++c) {
-      union {
-        bf16x8 v;
-        __bf16 e[8];
-      } a;
+    for (int sub = 0; sub < 2; ++sub) {
 #pragma unroll
-      for (int e = 0; e < 8; ++e)
-        a.e[e] = __bf16(dsv[2 * c + (e >> 2)][e & 3]);
-      dsfrag[c] = a.v;
+      for (int c = 0; c < 2; ++c) {
+        union {
+          bf16x8 v;
+          __bf16 e[8];
+        } a;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          a.e[e] = __bf16(dsv[sub][2 * c + (e >> 2)][e & 3]);
+        dsfrag[sub][c] = a.v;
+      }
     }
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
       const __bf16* kt_row = &Kt_lds[(n * 16 + li) * kStride];
-      acc_dq[n] = MFMA16(dsfrag[0], frag_row(kt_row, 0, g), acc_dq[n]);
-      acc_dq[n] = MFMA16(dsfrag[1], frag_row(kt_row, 32, g), acc_dq[n]);
+      const bf16x8 ktf0 = frag_row(kt_row, 0, g);
+      const bf16x8 ktf1 = frag_row(kt_row, 32, g);
+      acc_dq[0][n] = MFMA16(dsfrag[0][0], ktf0, acc_dq[0][n]);
+      acc_dq[1][n] = MFMA16(dsfrag[1][0], ktf0, acc_dq[1][n]);
+      acc_dq[0][n] = MFMA16(dsfrag[0][1], ktf1, acc_dq[0][n]);
+      acc_dq[1][n] = MFMA16(dsfrag[1][1], ktf1, acc_dq[1][n]);
     }
   }
 
   // epilogue: one store per element into dqkv Q slots
 #pragma unroll
-  for (int n = 0; n < 4; ++n) {
+  for (int sub = 0; sub < 2; ++sub) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int qr = q0 + wave * 16 + g * 4 + r;
-      if (qr < S) {
-        dqkv[(static_cast<int64_t>(b) * S + qr) * rs3 + h * 64 + n * 16 + li] =
-            __bf16(acc_dq[n][r]);
+    for (int n = 0; n < 4; ++n) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qr = q0 + wave * 32 + sub * 16 + g * 4 + r;
+        if (qr < S) {
+          dqkv[(static_cast<int64_t>(b) * S + qr) * rs3 + h * 64 + n * 16 +
+               li] = __bf16(acc_dq[sub][n][r]);
+        }
       }
     }
   }
@@ -744,12 +775,13 @@ torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
                      delta.data_ptr<float>(), B, S, NH);
 
   dim3 grid((S + 63) / 64, B * NH), block(256);
+  dim3 grid_dq((S + 127) / 128, B * NH);  // 128 q-rows per dq block
   const size_t lds = 6 * 64 * kStride * sizeof(__bf16) + 2 * 64 * sizeof(float);
   const size_t lds_dq = 3 * 64 * kStride * sizeof(__bf16);
   const float scale = 1.0f / sqrtf(64.f);
   const bool train_drop = p > 0.0;
-  auto args = [&](auto kernel, size_t lds_bytes) {
-    hipLaunchKernelGGL(kernel, grid, block, lds_bytes, stream,
+  auto args = [&](auto kernel, dim3 g, size_t lds_bytes) {
+    hipLaunchKernelGGL(kernel, g, block, lds_bytes, stream,
                        reinterpret_cast<const __bf16*>(dout_c.data_ptr()),
                        reinterpret_cast<const __bf16*>(qkv.data_ptr()),
                        seql.data_ptr<int>(), lse.data_ptr<float>(),
@@ -760,11 +792,11 @@ torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
                        static_cast<uint64_t>(offset));
   };
   if (train_drop) {
-    args(attn_bwd_kernel<true>, lds);
-    args(attn_dq_kernel<true>, lds_dq);
+    args(attn_bwd_kernel<true>, grid, lds);
+    args(attn_dq_kernel<true>, grid_dq, lds_dq);
   } else {
-    args(attn_bwd_kernel<false>, lds);
-    args(attn_dq_kernel<false>, lds_dq);
+    args(attn_bwd_kernel<false>, grid, lds);
+    args(attn_dq_kernel<false>, grid_dq, lds_dq);
   }
   return dqkv;
 }
