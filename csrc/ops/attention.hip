@@ -512,9 +512,12 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     const int* __restrict__ seqlens, const float* __restrict__ lse,
     const float* __restrict__ delta, __bf16* __restrict__ dqkv, int B, int S,
     int NH, float p, float scale, uint64_t seed, uint64_t offset) {
+  // 4 waves x 32 keys (two 16-key subtiles per wave): the heavy per-
+  // q-tile staging of Q/Q^T/dO/dO^T is amortized over 128 keys, and
+  // every staged Q/dO/Q^T/dO^T fragment feeds two MFMA chains.
   const int bh = blockIdx.y;
   const int b = bh / NH, h = bh % NH;
-  const int k0 = blockIdx.x * 64;
+  const int k0 = blockIdx.x * 128;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
   const int g = (lane >> 4), li = lane & 15;
@@ -531,18 +534,19 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   const uint64_t drop_base = offset + static_cast<uint64_t>(bh) * s2 * s2;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  __bf16* K_lds = reinterpret_cast<__bf16*>(smem);   // [64][72] natural
-  __bf16* V_lds = K_lds + 64 * kStride;              // [64][72] natural
-  __bf16* Q_lds = V_lds + 64 * kStride;              // [64][72] natural
+  __bf16* K_lds = reinterpret_cast<__bf16*>(smem);   // [128][72] natural
+  __bf16* V_lds = K_lds + 128 * kStride;             // [128][72] natural
+  __bf16* Q_lds = V_lds + 128 * kStride;             // [64][72] natural
   __bf16* Qt_lds = Q_lds + 64 * kStride;             // [64][72] transposed
   __bf16* dO_lds = Qt_lds + 64 * kStride;            // [64][72] natural
   __bf16* dOt_lds = dO_lds + 64 * kStride;           // [64][72] transposed
   float* lse_lds = reinterpret_cast<float*>(dOt_lds + 64 * kStride);  // [64]
   float* dlt_lds = lse_lds + 64;                                      // [64]
 
-  // stage K and V (natural) once
-  {
-    const int row = tid >> 2, colc = (tid & 3) * 16;
+  // stage K and V (natural) once: two 64-row passes
+#pragma unroll
+  for (int pass = 0; pass < 2; ++pass) {
+    const int row = pass * 64 + (tid >> 2), colc = (tid & 3) * 16;
     const int krow = k0 + row;
     __bf16 kv[16], vv[16];
     if (krow < S) {
@@ -569,8 +573,8 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   }
   __syncthreads();
 
-  f32x4 dv_acc[4] = {};  // dV^T[dh-tile][wave's 16 keys]
-  f32x4 dk_acc[4] = {};  // dK^T[d-tile][wave's 16 keys]
+  f32x4 dv_acc[2][4] = {};  // dV^T[sub][dh-tile] for the wave's 2x16 keys
+  f32x4 dk_acc[2][4] = {};
 
   const int n_q = (S + 63) / 64;
   for (int qt = 0; qt < n_q; ++qt) {
@@ -615,94 +619,115 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     }
     __syncthreads();
 
-    // recompute S tiles for this wave's 16 keys x 64 q-rows:
-    // S[q][key]: A = Q rows, B = K^T (natural K rows); C col = key = li
-    const int key_local = wave * 16 + li;
-    const int key_abs = k0 + key_local;
-    const __bf16* krow_n = &K_lds[key_local * kStride];
-    float pv[4][4];   // P[qtile mq][reg r] for this lane's key
-    float dsv[4][4];  // dS
+    // recompute S tiles for the wave's 2x16 keys x 64 q-rows:
+    // S[q][key]: A = Q rows, B = K^T (natural K rows); C col = key = li.
+    // Q/dO fragments are shared across the two key subtiles.
+    const int key_local0 = wave * 32 + li;
+    const int key_local1 = wave * 32 + 16 + li;
+    float pv[2][4][4];   // P[sub][qtile mq][reg r] for this lane's keys
+    float dsv[2][4][4];  // dS
 #pragma unroll
     for (int mq = 0; mq < 4; ++mq) {
       const __bf16* qrow_n = &Q_lds[(mq * 16 + li) * kStride];
-      f32x4 acc = {};
-      acc = MFMA16(frag_row(qrow_n, 0, g), frag_row(krow_n, 0, g), acc);
-      acc = MFMA16(frag_row(qrow_n, 32, g), frag_row(krow_n, 32, g), acc);
-      // dP tile: A = dO rows, B = V^T == natural V rows
       const __bf16* dorow = &dO_lds[(mq * 16 + li) * kStride];
-      const __bf16* vrow = &V_lds[key_local * kStride];
-      f32x4 dp = {};
-      dp = MFMA16(frag_row(dorow, 0, g), frag_row(vrow, 0, g), dp);
-      dp = MFMA16(frag_row(dorow, 32, g), frag_row(vrow, 32, g), dp);
-      const bool kvalid = key_abs < slen && key_abs < S;
-      // one Philox call serves a consecutive q pair (q_abs>>1 shared)
-      uint32_t r4[4];
+      const bf16x8 qa0 = frag_row(qrow_n, 0, g), qa1 = frag_row(qrow_n, 32, g);
+      const bf16x8 da0 = frag_row(dorow, 0, g), da1 = frag_row(dorow, 32, g);
+      f32x4 acc[2] = {}, dp[2] = {};
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int q_abs = q0 + mq * 16 + g * 4 + r;
-        const float row_lse = lse_lds[mq * 16 + g * 4 + r];
-        const float d_row = dlt_lds[mq * 16 + g * 4 + r];
-        float pr = (kvalid && q_abs < S)
-                       ? __expf(acc[r] * scale - row_lse)
-                       : 0.f;
-        float dpd = dp[r];
-        if (TRAIN_DROP) {
-          if ((r & 1) == 0) {
-            philox(drop_base + static_cast<uint64_t>(q_abs >> 1) * s2 +
-                       (key_abs >> 1),
-                   r4);
+      for (int sub = 0; sub < 2; ++sub) {
+        const int key_local = sub ? key_local1 : key_local0;
+        const __bf16* krow_n = &K_lds[key_local * kStride];
+        const __bf16* vrow = &V_lds[key_local * kStride];
+        acc[sub] = MFMA16(qa0, frag_row(krow_n, 0, g), acc[sub]);
+        acc[sub] = MFMA16(qa1, frag_row(krow_n, 32, g), acc[sub]);
+        dp[sub] = MFMA16(da0, frag_row(vrow, 0, g), dp[sub]);
+        dp[sub] = MFMA16(da1, frag_row(vrow, 32, g), dp[sub]);
+      }
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+        const int key_abs = k0 + (sub ? key_local1 : key_local0);
+        const bool kvalid = key_abs < slen && key_abs < S;
+        // one Philox call serves a consecutive q pair (q_abs>>1 shared)
+        uint32_t r4[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int q_abs = q0 + mq * 16 + g * 4 + r;
+          const float row_lse = lse_lds[mq * 16 + g * 4 + r];
+          const float d_row = dlt_lds[mq * 16 + g * 4 + r];
+          float pr = (kvalid && q_abs < S)
+                         ? __expf(acc[sub][r] * scale - row_lse)
+                         : 0.f;
+          float dpd = dp[sub][r];
+          if (TRAIN_DROP) {
+            if ((r & 1) == 0) {
+              philox(drop_base + static_cast<uint64_t>(q_abs >> 1) * s2 +
+                         (key_abs >> 1),
+                     r4);
+            }
+            const bool keep =
+                u32_to_uniform(r4[(q_abs & 1) * 2 + (key_abs & 1)]) >= p;
+            dpd = keep ? dpd * inv_keep : 0.f;
+            pv[sub][mq][r] = keep ? pr * inv_keep : 0.f;
+          } else {
+            pv[sub][mq][r] = pr;
           }
-          const bool keep =
-              u32_to_uniform(r4[(q_abs & 1) * 2 + (key_abs & 1)]) >= p;
-          dpd = keep ? dpd * inv_keep : 0.f;
-          pv[mq][r] = keep ? pr * inv_keep : 0.f;  // dropped P feeds dV
-        } else {
-          pv[mq][r] = pr;
+          dsv[sub][mq][r] = pr * (dpd - d_row) * scale;
         }
-        dsv[mq][r] = pr * (dpd - d_row) * scale;
       }
     }
 
     // chain P and dS registers into B-fragments over q (chunk c: q 32c+..)
-    bf16x8 pfrag[2], dsfrag[2];
+    bf16x8 pfrag[2][2], dsfrag[2][2];
 #pragma unroll
-    for (int c = 0; c < 2; ++c) {
-      union {
-        bf16x8 v;
-        __bf16 e[8];
-      } a, d2;
+    for (int sub = 0; sub < 2; ++sub) {
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        a.e[e] = __bf16(pv[2 * c + (e >> 2)][e & 3]);
-        d2.e[e] = __bf16(dsv[2 * c + (e >> 2)][e & 3]);
+      for (int c = 0; c < 2; ++c) {
+        union {
+          bf16x8 v;
+          __bf16 e[8];
+        } a, d2;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          a.e[e] = __bf16(pv[sub][2 * c + (e >> 2)][e & 3]);
+          d2.e[e] = __bf16(dsv[sub][2 * c + (e >> 2)][e & 3]);
+        }
+        pfrag[sub][c] = a.v;
+        dsfrag[sub][c] = d2.v;
       }
-      pfrag[c] = a.v;
-      dsfrag[c] = d2.v;
     }
 
-    // dV^T += dO^T x P ; dK^T += Q^T x dS  (per dh/d tile m)
+    // dV^T += dO^T x P ; dK^T += Q^T x dS — dO^T/Q^T fragments shared
 #pragma unroll
     for (int m = 0; m < 4; ++m) {
       const __bf16* dot_row = &dOt_lds[(m * 16 + li) * kStride];
       const __bf16* qt_row = &Qt_lds[(m * 16 + li) * kStride];
-      dv_acc[m] = MFMA16(frag_row(dot_row, 0, g), pfrag[0], dv_acc[m]);
-      dv_acc[m] = MFMA16(frag_row(dot_row, 32, g), pfrag[1], dv_acc[m]);
-      dk_acc[m] = MFMA16(frag_row(qt_row, 0, g), dsfrag[0], dk_acc[m]);
-      dk_acc[m] = MFMA16(frag_row(qt_row, 32, g), dsfrag[1], dk_acc[m]);
+      const bf16x8 dt0 = frag_row(dot_row, 0, g), dt1 = frag_row(dot_row, 32, g);
+      const bf16x8 qt0 = frag_row(qt_row, 0, g), qt1 = frag_row(qt_row, 32, g);
+      dv_acc[0][m] = MFMA16(dt0, pfrag[0][0], dv_acc[0][m]);
+      dv_acc[1][m] = MFMA16(dt0, pfrag[1][0], dv_acc[1][m]);
+      dv_acc[0][m] = MFMA16(dt1, pfrag[0][1], dv_acc[0][m]);
+      dv_acc[1][m] = MFMA16(dt1, pfrag[1][1], dv_acc[1][m]);
+      dk_acc[0][m] = MFMA16(qt0, dsfrag[0][0], dk_acc[0][m]);
+      dk_acc[1][m] = MFMA16(qt0, dsfrag[1][0], dk_acc[1][m]);
+      dk_acc[0][m] = MFMA16(qt1, dsfrag[0][1], dk_acc[0][m]);
+      dk_acc[1][m] = MFMA16(qt1, dsfrag[1][1], dk_acc[1][m]);
     }
   }
 
-  // write dK/dV straight into dqkv (this block owns keys k0..k0+63)
+  // write dK/dV straight into dqkv (this block owns keys k0..k0+127)
 #pragma unroll
-  for (int m = 0; m < 4; ++m) {
+  for (int sub = 0; sub < 2; ++sub) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int dh = m * 16 + g * 4 + r;  // C row = dh/d index
-      const int key_abs = k0 + wave * 16 + li;
-      if (key_abs < S) {
-        const int64_t rowb = static_cast<int64_t>(b) * S + key_abs;
-        dqkv[rowb * rs3 + H + h * 64 + dh] = __bf16(dk_acc[m][r]);
-        dqkv[rowb * rs3 + 2 * H + h * 64 + dh] = __bf16(dv_acc[m][r]);
+    for (int m = 0; m < 4; ++m) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int dh = m * 16 + g * 4 + r;  // C row = dh/d index
+        const int key_abs = k0 + wave * 32 + sub * 16 + li;
+        if (key_abs < S) {
+          const int64_t rowb = static_cast<int64_t>(b) * S + key_abs;
+          dqkv[rowb * rs3 + H + h * 64 + dh] = __bf16(dk_acc[sub][m][r]);
+          dqkv[rowb * rs3 + 2 * H + h * 64 + dh] = __bf16(dv_acc[sub][m][r]);
+        }
       }
     }
   }
@@ -774,10 +799,18 @@ torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
                      reinterpret_cast<const __bf16*>(out.data_ptr()),
                      delta.data_ptr<float>(), B, S, NH);
 
-  dim3 grid((S + 63) / 64, B * NH), block(256);
-  dim3 grid_dq((S + 127) / 128, B * NH);  // 128 q-rows per dq block
-  const size_t lds = 6 * 64 * kStride * sizeof(__bf16) + 2 * 64 * sizeof(float);
+  dim3 grid((S + 127) / 128, B * NH), block(256);  // 128 keys per bwd block
+  dim3 grid_dq((S + 127) / 128, B * NH);           // 128 q-rows per dq block
+  // K/V hold 128 rows; Q/Qt/dO/dOt 64 each -> ~74 KB (over the 64 KB
+  // default dynamic-LDS cap; MI355X has 160 KB per CU)
+  const size_t lds = 8 * 64 * kStride * sizeof(__bf16) + 2 * 64 * sizeof(float);
   const size_t lds_dq = 3 * 64 * kStride * sizeof(__bf16);
+  HIP_CHECK(hipFuncSetAttribute(
+      reinterpret_cast<const void*>(&attn_bwd_kernel<true>),
+      hipFuncAttributeMaxDynamicSharedMemorySize, lds));
+  HIP_CHECK(hipFuncSetAttribute(
+      reinterpret_cast<const void*>(&attn_bwd_kernel<false>),
+      hipFuncAttributeMaxDynamicSharedMemorySize, lds));
   const float scale = 1.0f / sqrtf(64.f);
   const bool train_drop = p > 0.0;
   auto args = [&](auto kernel, dim3 g, size_t lds_bytes) {
