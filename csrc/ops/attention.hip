@@ -59,13 +59,6 @@ __device__ __forceinline__ bf16x8 frag_row(const __bf16* row, int base,
   return r.v;
 }
 
-__device__ __forceinline__ float uniform_at(const Philox& ph, uint64_t ctr,
-                                            int word) {
-  uint32_t r4[4];
-  ph(ctr, r4);
-  return u32_to_uniform(r4[word]);
-}
-
 // ---------------------------------------------------------------------------
 // forward
 // ---------------------------------------------------------------------------
