@@ -59,6 +59,9 @@ def parse_args():
     p.add_argument("--local_batch", type=int, default=0)
     p.add_argument("--model_config", type=str,
                    default="config/bert_large_uncased_config.json")
+    p.add_argument("--grad_compress", type=str, default=None,
+                   choices=["bf16", "fp16"],
+                   help="compressed-gradient all-reduce comm hook")
     p.add_argument("--window_autocast", action="store_true",
                    help="hold one autocast region per accumulation window "
                         "so weight-cast caching spans its micro-steps")
@@ -111,7 +114,8 @@ def main():
     if args.pure_bf16:
         model = model.to(torch.bfloat16)
     criterion = BertPretrainingCriterion(config.vocab_size)
-    model = comm.wrap_ddp(model, local_rank)
+    model = comm.wrap_ddp(model, local_rank,
+                          grad_compress=args.grad_compress)
     named = list(model.named_parameters())
     no_decay = ("bias", "LayerNorm", "qkv_bias")
     optimizer = FusedLAMB(

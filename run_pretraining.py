@@ -83,6 +83,9 @@ def parse_arguments(args=None) -> argparse.Namespace:
                         help="pad vocab for MFMA tile alignment "
                              "(reference pads to 8: run_pretraining.py:237)")
     parser.add_argument("--num_workers", type=int, default=4)
+    parser.add_argument("--grad_compress", type=str, default=None,
+                        choices=["bf16", "fp16"],
+                        help="compressed-gradient all-reduce comm hook")
     parser.add_argument("--kfac", action="store_true")
     parser.add_argument("--kfac_inv_interval", type=int, default=10)
     parser.add_argument("--kfac_factor_interval", type=int, default=1)
@@ -175,7 +178,8 @@ def prepare_model(args, device):
     if args.checkpoint_activations:
         model.checkpoint_activations(True)
     global_steps = max(0, resume_step - args.previous_phase_end_step)
-    model = comm.wrap_ddp(model, args.local_rank)
+    model = comm.wrap_ddp(model, args.local_rank,
+                          grad_compress=args.grad_compress)
     return model, config, resume_state, global_steps
 
 

@@ -150,3 +150,49 @@ def test_sharded_dataset_multirank_workers(tmp_path):
         p.join(timeout=30)
     for rank, count in results:
         assert count == 16, f"rank {rank}: {count}"  # 32 samples/rank / 4 * 2 epochs
+
+
+def _run_compress_hook(rank, world, port, q):
+    try:
+        _init(rank, world, port + 40)
+        torch.manual_seed(0)
+        lin = torch.nn.Linear(32, 32)
+        from bert_pytorch_amd.parallel import comm as bpa_comm
+
+        ddp = bpa_comm.wrap_ddp(lin, 0, grad_compress="fp16")
+        g = torch.Generator().manual_seed(200 + rank)
+        x = torch.randn(8, 32, generator=g)
+        ddp(x).sum().backward()
+        # expected: fp16-rounded average of per-rank grads
+        solo = torch.nn.Linear(32, 32)
+        solo.load_state_dict(lin.state_dict())
+        solo(x).sum().backward()
+        manual = solo.weight.grad.clone()
+        dist.all_reduce(manual)
+        manual /= world
+        ok = torch.allclose(
+            lin.weight.grad, manual, rtol=2e-3, atol=2e-3
+        )
+        q.put((rank, bool(ok)))
+        dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"ERROR: {e}"))
+
+
+@pytest.mark.timeout(120)
+def test_grad_compress_hook_gloo():
+    """fp16-compressed all-reduce comm hook matches the fp32 average
+    within half-precision tolerance (2 ranks, gloo)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [
+        ctx.Process(target=_run_compress_hook, args=(r, WORLD, 29551, q))
+        for r in range(WORLD)
+    ]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=110) for _ in range(WORLD)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, ok in results:
+        assert ok is True, f"rank {rank}: {ok}"
