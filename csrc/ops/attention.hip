@@ -411,8 +411,14 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
     }
     __syncthreads();
 
-    // S^T and dP^T tiles (C[key][q], q = li); K/V fragments shared
-    float dsv[2][4][4];
+    // S^T and dP^T tiles (C[key][q], q = li); K/V fragments shared.
+    // dS packs straight into its A-fragment slots (element
+    // e = (t&1)*4 + r of chunk t>>1) - no [2][4][4] staging array.
+    union FB {
+      bf16x8 v;
+      __bf16 e[8];
+    };
+    FB dsfrag[2][2];
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
       const __bf16* krow = &K_lds[(t * 16 + li) * kStride];
@@ -449,25 +455,9 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
                 u32_to_uniform(r4[(q_row[sub] & 1) * 2 + (key & 1)]) >= p;
             dpd = keep ? dpd * inv_keep : 0.f;
           }
-          dsv[sub][t][r] = pr * (dpd - dlt_q[sub]) * scale;
+          dsfrag[sub][t >> 1].e[(t & 1) * 4 + r] =
+              __bf16(pr * (dpd - dlt_q[sub]) * scale);
         }
-      }
-    }
-
-    // chain dS^T into A-fragments over keys; K^T fragments shared
-    bf16x8 dsfrag[2][2];
-#pragma unroll
-    for (int sub = 0; sub < 2; ++sub) {
-#pragma unroll
-      for (int c = 0; c < 2; ++c) {
-        union {
-          bf16x8 v;
-          __bf16 e[8];
-        } a;
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-          a.e[e] = __bf16(dsv[sub][2 * c + (e >> 2)][e & 3]);
-        dsfrag[sub][c] = a.v;
       }
     }
 #pragma unroll
@@ -475,10 +465,10 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
       const __bf16* kt_row = &Kt_lds[(n * 16 + li) * kStride];
       const bf16x8 ktf0 = frag_row(kt_row, 0, g);
       const bf16x8 ktf1 = frag_row(kt_row, 32, g);
-      acc_dq[0][n] = MFMA16(dsfrag[0][0], ktf0, acc_dq[0][n]);
-      acc_dq[1][n] = MFMA16(dsfrag[1][0], ktf0, acc_dq[1][n]);
-      acc_dq[0][n] = MFMA16(dsfrag[0][1], ktf1, acc_dq[0][n]);
-      acc_dq[1][n] = MFMA16(dsfrag[1][1], ktf1, acc_dq[1][n]);
+      acc_dq[0][n] = MFMA16(dsfrag[0][0].v, ktf0, acc_dq[0][n]);
+      acc_dq[1][n] = MFMA16(dsfrag[1][0].v, ktf0, acc_dq[1][n]);
+      acc_dq[0][n] = MFMA16(dsfrag[0][1].v, ktf1, acc_dq[0][n]);
+      acc_dq[1][n] = MFMA16(dsfrag[1][1].v, ktf1, acc_dq[1][n]);
     }
   }
 
@@ -617,8 +607,15 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     // Q/dO fragments are shared across the two key subtiles.
     const int key_local0 = wave * 32 + li;
     const int key_local1 = wave * 32 + 16 + li;
-    float pv[2][4][4];   // P[sub][qtile mq][reg r] for this lane's keys
-    float dsv[2][4][4];  // dS
+    // P and dS go STRAIGHT into their B-fragment slots as each q-tile
+    // is produced (element e = (mq&1)*4 + r of chunk mq>>1) - no
+    // [2][4][4] staging arrays (they cost ~64 VGPRs and halved
+    // occupancy at 324 VGPRs/wave)
+    union FB {
+      bf16x8 v;
+      __bf16 e[8];
+    };
+    FB pfrag[2][2], dsfrag[2][2];
 #pragma unroll
     for (int mq = 0; mq < 4; ++mq) {
       const __bf16* qrow_n = &Q_lds[(mq * 16 + li) * kStride];
@@ -651,6 +648,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
                          ? __expf(acc[sub][r] * scale - row_lse)
                          : 0.f;
           float dpd = dp[sub][r];
+          float pkeep = pr;
           if (TRAIN_DROP) {
             if ((r & 1) == 0) {
               philox(drop_base + static_cast<uint64_t>(q_abs >> 1) * s2 +
@@ -660,32 +658,12 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
             const bool keep =
                 u32_to_uniform(r4[(q_abs & 1) * 2 + (key_abs & 1)]) >= p;
             dpd = keep ? dpd * inv_keep : 0.f;
-            pv[sub][mq][r] = keep ? pr * inv_keep : 0.f;
-          } else {
-            pv[sub][mq][r] = pr;
+            pkeep = keep ? pr * inv_keep : 0.f;  // dropped P feeds dV
           }
-          dsv[sub][mq][r] = pr * (dpd - d_row) * scale;
+          pfrag[sub][mq >> 1].e[(mq & 1) * 4 + r] = __bf16(pkeep);
+          dsfrag[sub][mq >> 1].e[(mq & 1) * 4 + r] =
+              __bf16(pr * (dpd - d_row) * scale);
         }
-      }
-    }
-
-    // chain P and dS registers into B-fragments over q (chunk c: q 32c+..)
-    bf16x8 pfrag[2][2], dsfrag[2][2];
-#pragma unroll
-    for (int sub = 0; sub < 2; ++sub) {
-#pragma unroll
-      for (int c = 0; c < 2; ++c) {
-        union {
-          bf16x8 v;
-          __bf16 e[8];
-        } a, d2;
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          a.e[e] = __bf16(pv[sub][2 * c + (e >> 2)][e & 3]);
-          d2.e[e] = __bf16(dsv[sub][2 * c + (e >> 2)][e & 3]);
-        }
-        pfrag[sub][c] = a.v;
-        dsfrag[sub][c] = d2.v;
       }
     }
 
@@ -696,14 +674,14 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       const __bf16* qt_row = &Qt_lds[(m * 16 + li) * kStride];
       const bf16x8 dt0 = frag_row(dot_row, 0, g), dt1 = frag_row(dot_row, 32, g);
       const bf16x8 qt0 = frag_row(qt_row, 0, g), qt1 = frag_row(qt_row, 32, g);
-      dv_acc[0][m] = MFMA16(dt0, pfrag[0][0], dv_acc[0][m]);
-      dv_acc[1][m] = MFMA16(dt0, pfrag[1][0], dv_acc[1][m]);
-      dv_acc[0][m] = MFMA16(dt1, pfrag[0][1], dv_acc[0][m]);
-      dv_acc[1][m] = MFMA16(dt1, pfrag[1][1], dv_acc[1][m]);
-      dk_acc[0][m] = MFMA16(qt0, dsfrag[0][0], dk_acc[0][m]);
-      dk_acc[1][m] = MFMA16(qt0, dsfrag[1][0], dk_acc[1][m]);
-      dk_acc[0][m] = MFMA16(qt1, dsfrag[0][1], dk_acc[0][m]);
-      dk_acc[1][m] = MFMA16(qt1, dsfrag[1][1], dk_acc[1][m]);
+      dv_acc[0][m] = MFMA16(dt0, pfrag[0][0].v, dv_acc[0][m]);
+      dv_acc[1][m] = MFMA16(dt0, pfrag[1][0].v, dv_acc[1][m]);
+      dv_acc[0][m] = MFMA16(dt1, pfrag[0][1].v, dv_acc[0][m]);
+      dv_acc[1][m] = MFMA16(dt1, pfrag[1][1].v, dv_acc[1][m]);
+      dk_acc[0][m] = MFMA16(qt0, dsfrag[0][0].v, dk_acc[0][m]);
+      dk_acc[1][m] = MFMA16(qt0, dsfrag[1][0].v, dk_acc[1][m]);
+      dk_acc[0][m] = MFMA16(qt1, dsfrag[0][1].v, dk_acc[0][m]);
+      dk_acc[1][m] = MFMA16(qt1, dsfrag[1][1].v, dk_acc[1][m]);
     }
   }
 
