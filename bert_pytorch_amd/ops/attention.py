@@ -27,18 +27,21 @@ class _FusedAttention(torch.autograd.Function):
         seed, offset = (
             next_philox(bsz * num_heads * seq * seq) if p_eff > 0 else (0, 0)
         )
-        out, lse = ext.attention_fwd(qkv, seqlens, num_heads, p_eff, seed, offset)
-        ctx.save_for_backward(qkv, seqlens, out, lse)
+        out, lse, dmask = ext.attention_fwd(
+            qkv, seqlens, num_heads, p_eff, seed, offset
+        )
+        ctx.save_for_backward(qkv, seqlens, out, lse, dmask)
         ctx.meta = (num_heads, p_eff, seed, offset)
         return out
 
     @staticmethod
     def backward(ctx, dout):
         ext = extension()
-        qkv, seqlens, out, lse = ctx.saved_tensors
+        qkv, seqlens, out, lse, dmask = ctx.saved_tensors
         num_heads, p_eff, seed, offset = ctx.meta
         dqkv = ext.attention_bwd(
-            dout.contiguous(), qkv, seqlens, out, lse, num_heads, p_eff, seed, offset
+            dout.contiguous(), qkv, seqlens, out, lse, dmask, num_heads,
+            p_eff, seed, offset
         )
         return dqkv, None, None, None, None
 

@@ -40,8 +40,9 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor qkv,
                                          int64_t seed, int64_t offset);
 torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
                             torch::Tensor seqlens, torch::Tensor out,
-                            torch::Tensor lse, int64_t num_heads, double p,
-                            int64_t seed, int64_t offset);
+                            torch::Tensor lse, torch::Tensor dmask,
+                            int64_t num_heads, double p, int64_t seed,
+                            int64_t offset);
 torch::Tensor multi_tensor_l2norm_sq(std::vector<torch::Tensor> tensors);
 void multi_tensor_clip_scale(std::vector<torch::Tensor> grads,
                              torch::Tensor gnorm_sq, double max_norm);

@@ -65,8 +65,9 @@ __device__ __forceinline__ bf16x8 frag_row(const __bf16* row, int base,
 template <bool TRAIN_DROP>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const __bf16* __restrict__ qkv, const int* __restrict__ seqlens,
-    __bf16* __restrict__ out, float* __restrict__ lse_out, int B, int S,
-    int NH, float p, float scale, uint64_t seed, uint64_t offset) {
+    __bf16* __restrict__ out, float* __restrict__ lse_out,
+    uint8_t* __restrict__ dmask, int B, int S, int NH, float p, float scale,
+    uint64_t seed, uint64_t offset) {
   // 4 waves x 32 q-rows (two 16-row subtiles per wave): the K/V tile is
   // staged once per 128 q-rows, each K/V fragment read feeds TWO
   // independent MFMA chains, and the softmax work of the two subtiles
@@ -85,9 +86,11 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int slen = seqlens[b];
   const float inv_keep = TRAIN_DROP ? 1.f / (1.f - p) : 1.f;
   Philox philox(seed);
-  const int s2 = S >> 1;
+  const int s4 = S >> 2;
   const uint64_t drop_base =
-      offset + static_cast<uint64_t>(bh) * s2 * s2;
+      offset + static_cast<uint64_t>(bh) * S * s4;
+  uint8_t* mask_base =
+      TRAIN_DROP ? dmask + static_cast<int64_t>(bh) * S * S : nullptr;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* K_lds = reinterpret_cast<__bf16*>(smem);             // [64][72]
@@ -200,25 +203,31 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
           acc_o[sub][n][r] *= alpha_lds[(wave * 2 + sub) * 16 + g * 4 + r];
 
     if (TRAIN_DROP) {
+      // one Philox call per key-quad (all 4 words used by THIS lane);
+      // the keep-mask is stored as bytes so the backward kernels read
+      // it instead of regenerating (dropout RNG was ~2x fwd cost and
+      // similar in bwd/dq)
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {
         const int q_abs = q_row[sub];
+        const bool q_ok = q_abs < S;
 #pragma unroll
         for (int t = 0; t < 4; ++t) {
+          const int key0 = k0 + t * 16 + g * 4;
+          uint32_t r4[4];
+          philox(drop_base + static_cast<uint64_t>(q_abs) * s4 + (key0 >> 2),
+                 r4);
+          uint8_t mb[4];
 #pragma unroll
-          for (int rpair = 0; rpair < 2; ++rpair) {
-            const int key = k0 + t * 16 + g * 4 + rpair * 2;
-            uint32_t r4[4];
-            philox(drop_base +
-                       static_cast<uint64_t>(q_abs >> 1) * s2 + (key >> 1),
-                   r4);
-            const int w0 = (q_abs & 1) * 2;
-#pragma unroll
-            for (int j = 0; j < 2; ++j) {
-              const bool keep = u32_to_uniform(r4[w0 + j]) >= p;
-              sv[sub][t][rpair * 2 + j] =
-                  keep ? sv[sub][t][rpair * 2 + j] * inv_keep : 0.f;
-            }
+          for (int j = 0; j < 4; ++j) {
+            const bool keep = u32_to_uniform(r4[j]) >= p;
+            mb[j] = keep ? 1 : 0;
+            sv[sub][t][j] = keep ? sv[sub][t][j] * inv_keep : 0.f;
+          }
+          if (q_ok) {
+            *reinterpret_cast<uint32_t*>(
+                mask_base + static_cast<int64_t>(q_abs) * S + key0) =
+                *reinterpret_cast<const uint32_t*>(mb);
           }
         }
       }
@@ -327,8 +336,9 @@ template <bool TRAIN_DROP>
 __global__ __launch_bounds__(256) void attn_dq_kernel(
     const __bf16* __restrict__ dout, const __bf16* __restrict__ qkv,
     const int* __restrict__ seqlens, const float* __restrict__ lse,
-    const float* __restrict__ delta, __bf16* __restrict__ dqkv, int B, int S,
-    int NH, float p, float scale, uint64_t seed, uint64_t offset) {
+    const float* __restrict__ delta, const uint8_t* __restrict__ dmask,
+    __bf16* __restrict__ dqkv, int B, int S, int NH, float p, float scale,
+    uint64_t seed, uint64_t offset) {
   // 4 waves x 32 q-rows (two 16-row subtiles per wave), mirroring
   // attn_fwd_kernel: K/V/K^T staged once per 128 q-rows, every staged
   // fragment feeds two independent MFMA chains.
@@ -346,9 +356,8 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
   const __bf16* dobase = dout + static_cast<int64_t>(b) * S * H + h * 64;
   const int slen = seqlens[b];
   const float inv_keep = TRAIN_DROP ? 1.f / (1.f - p) : 1.f;
-  Philox philox(seed);
-  const int s2 = S >> 1;
-  const uint64_t drop_base = offset + static_cast<uint64_t>(bh) * s2 * s2;
+  const uint8_t* mask_base =
+      TRAIN_DROP ? dmask + static_cast<int64_t>(bh) * S * S : nullptr;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* K_lds = reinterpret_cast<__bf16*>(smem);   // [64][72] natural
@@ -435,8 +444,13 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
       }
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {
-        // one Philox call serves a consecutive key pair (key>>1 shared)
-        uint32_t r4[4];
+        // read the forward's stored keep-mask: one 4-byte load per quad
+        uint8_t mb[4] = {1, 1, 1, 1};
+        if (TRAIN_DROP && q_row[sub] < S) {
+          *reinterpret_cast<uint32_t*>(mb) = *reinterpret_cast<const uint32_t*>(
+              mask_base + static_cast<int64_t>(q_row[sub]) * S + k0 + t * 16 +
+              g * 4);
+        }
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int key = k0 + t * 16 + g * 4 + r;
@@ -445,15 +459,7 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
               valid ? __expf(sacc[sub][r] * scale - lse_q[sub]) : 0.f;
           float dpd = dpacc[sub][r];
           if (TRAIN_DROP) {
-            if ((r & 1) == 0) {
-              philox(drop_base +
-                         static_cast<uint64_t>(q_row[sub] >> 1) * s2 +
-                         (key >> 1),
-                     r4);
-            }
-            const bool keep =
-                u32_to_uniform(r4[(q_row[sub] & 1) * 2 + (key & 1)]) >= p;
-            dpd = keep ? dpd * inv_keep : 0.f;
+            dpd = mb[r] ? dpd * inv_keep : 0.f;
           }
           dsfrag[sub][t >> 1].e[(t & 1) * 4 + r] =
               __bf16(pr * (dpd - dlt_q[sub]) * scale);
@@ -493,8 +499,9 @@ template <bool TRAIN_DROP>
 __global__ __launch_bounds__(256) void attn_bwd_kernel(
     const __bf16* __restrict__ dout, const __bf16* __restrict__ qkv,
     const int* __restrict__ seqlens, const float* __restrict__ lse,
-    const float* __restrict__ delta, __bf16* __restrict__ dqkv, int B, int S,
-    int NH, float p, float scale, uint64_t seed, uint64_t offset) {
+    const float* __restrict__ delta, const uint8_t* __restrict__ dmask,
+    __bf16* __restrict__ dqkv, int B, int S, int NH, float p, float scale,
+    uint64_t seed, uint64_t offset) {
   // 4 waves x 32 keys (two 16-key subtiles per wave): the heavy per-
   // q-tile staging of Q/Q^T/dO/dO^T is amortized over 128 keys, and
   // every staged Q/dO/Q^T/dO^T fragment feeds two MFMA chains.
@@ -512,9 +519,8 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   const __bf16* dobase = dout + static_cast<int64_t>(b) * S * H + h * 64;
   const int slen = seqlens[b];
   const float inv_keep = TRAIN_DROP ? 1.f / (1.f - p) : 1.f;
-  Philox philox(seed);
-  const int s2 = S >> 1;
-  const uint64_t drop_base = offset + static_cast<uint64_t>(bh) * s2 * s2;
+  const uint8_t* mask_base =
+      TRAIN_DROP ? dmask + static_cast<int64_t>(bh) * S * S : nullptr;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* K_lds = reinterpret_cast<__bf16*>(smem);   // [128][72] natural
@@ -525,6 +531,9 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   __bf16* dOt_lds = dO_lds + 64 * kStride;           // [64][72] transposed
   float* lse_lds = reinterpret_cast<float*>(dOt_lds + 64 * kStride);  // [64]
   float* dlt_lds = lse_lds + 64;                                      // [64]
+  // keep-mask tile [64 q][128 keys] staged per q-tile (the raw global
+  // reads are byte columns - latency-bound at this kernel's occupancy)
+  uint8_t* mk_lds = reinterpret_cast<uint8_t*>(dlt_lds + 64);
 
   // stage K and V (natural) once: two 64-row passes
 #pragma unroll
@@ -637,8 +646,6 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       for (int sub = 0; sub < 2; ++sub) {
         const int key_abs = k0 + (sub ? key_local1 : key_local0);
         const bool kvalid = key_abs < slen && key_abs < S;
-        // one Philox call serves a consecutive q pair (q_abs>>1 shared)
-        uint32_t r4[4];
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int q_abs = q0 + mq * 16 + g * 4 + r;
@@ -650,13 +657,9 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
           float dpd = dp[sub][r];
           float pkeep = pr;
           if (TRAIN_DROP) {
-            if ((r & 1) == 0) {
-              philox(drop_base + static_cast<uint64_t>(q_abs >> 1) * s2 +
-                         (key_abs >> 1),
-                     r4);
-            }
             const bool keep =
-                u32_to_uniform(r4[(q_abs & 1) * 2 + (key_abs & 1)]) >= p;
+                mk_lds[(mq * 16 + g * 4 + r) * 128 +
+                       (sub ? key_local1 : key_local0)] != 0;
             dpd = keep ? dpd * inv_keep : 0.f;
             pkeep = keep ? pr * inv_keep : 0.f;  // dropped P feeds dV
           }
@@ -722,36 +725,45 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor qkv,
   auto seql = seqlens.to(qkv.device(), torch::kInt32).contiguous();
   auto out = torch::empty({B, S, H}, qkv.options());
   auto lse = torch::empty({B * NH, S}, qkv.options().dtype(torch::kFloat32));
+  const bool train_drop = p > 0.0;
+  // keep-mask bytes [B*NH, S, S]: written once here, read (not
+  // regenerated) by both backward kernels
+  auto dmask = train_drop
+                   ? torch::empty({static_cast<int64_t>(B) * NH, S, S},
+                                  qkv.options().dtype(torch::kUInt8))
+                   : torch::empty({0}, qkv.options().dtype(torch::kUInt8));
   const float scale = 1.0f / sqrtf(64.f);
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid((S + 127) / 128, B * NH), block(256);  // 128 q-rows per block
   const size_t lds =
       2 * 64 * kStride * sizeof(__bf16) + 4 * 64 * sizeof(float);
-  const bool train_drop = p > 0.0;
   if (train_drop) {
     hipLaunchKernelGGL((attn_fwd_kernel<true>), grid, block, lds, stream,
                        reinterpret_cast<const __bf16*>(qkv.data_ptr()),
                        seql.data_ptr<int>(),
                        reinterpret_cast<__bf16*>(out.data_ptr()),
-                       lse.data_ptr<float>(), B, S, NH, static_cast<float>(p),
-                       scale, static_cast<uint64_t>(seed),
+                       lse.data_ptr<float>(), dmask.data_ptr<uint8_t>(), B, S,
+                       NH, static_cast<float>(p), scale,
+                       static_cast<uint64_t>(seed),
                        static_cast<uint64_t>(offset));
   } else {
     hipLaunchKernelGGL((attn_fwd_kernel<false>), grid, block, lds, stream,
                        reinterpret_cast<const __bf16*>(qkv.data_ptr()),
                        seql.data_ptr<int>(),
                        reinterpret_cast<__bf16*>(out.data_ptr()),
-                       lse.data_ptr<float>(), B, S, NH, static_cast<float>(p),
-                       scale, static_cast<uint64_t>(seed),
+                       lse.data_ptr<float>(), nullptr, B, S, NH,
+                       static_cast<float>(p), scale,
+                       static_cast<uint64_t>(seed),
                        static_cast<uint64_t>(offset));
   }
-  return {out, lse};
+  return {out, lse, dmask};
 }
 
 torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
                             torch::Tensor seqlens, torch::Tensor out,
-                            torch::Tensor lse, int64_t num_heads, double p,
-                            int64_t seed, int64_t offset) {
+                            torch::Tensor lse, torch::Tensor dmask,
+                            int64_t num_heads, double p, int64_t seed,
+                            int64_t offset) {
   const int B = qkv.size(0), S = qkv.size(1);
   const int H = qkv.size(2) / 3;
   const int NH = static_cast<int>(num_heads);
@@ -772,9 +784,10 @@ torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
 
   dim3 grid((S + 127) / 128, B * NH), block(256);  // 128 keys per bwd block
   dim3 grid_dq((S + 127) / 128, B * NH);           // 128 q-rows per dq block
-  // K/V hold 128 rows; Q/Qt/dO/dOt 64 each -> ~74 KB (over the 64 KB
-  // default dynamic-LDS cap; MI355X has 160 KB per CU)
-  const size_t lds = 8 * 64 * kStride * sizeof(__bf16) + 2 * 64 * sizeof(float);
+  // K/V hold 128 rows; Q/Qt/dO/dOt 64 each; + 8 KB mask tile -> ~82 KB
+  // (over the 64 KB default dynamic-LDS cap; MI355X has 160 KB per CU)
+  const size_t lds = 8 * 64 * kStride * sizeof(__bf16) +
+                     2 * 64 * sizeof(float) + 64 * 128;
   const size_t lds_dq = 3 * 64 * kStride * sizeof(__bf16);
   HIP_CHECK(hipFuncSetAttribute(
       reinterpret_cast<const void*>(&attn_bwd_kernel<true>),
@@ -784,12 +797,14 @@ torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
       hipFuncAttributeMaxDynamicSharedMemorySize, lds));
   const float scale = 1.0f / sqrtf(64.f);
   const bool train_drop = p > 0.0;
+  const uint8_t* mask_ptr =
+      train_drop ? dmask.data_ptr<uint8_t>() : nullptr;
   auto args = [&](auto kernel, dim3 g, size_t lds_bytes) {
     hipLaunchKernelGGL(kernel, g, block, lds_bytes, stream,
                        reinterpret_cast<const __bf16*>(dout_c.data_ptr()),
                        reinterpret_cast<const __bf16*>(qkv.data_ptr()),
                        seql.data_ptr<int>(), lse.data_ptr<float>(),
-                       delta.data_ptr<float>(),
+                       delta.data_ptr<float>(), mask_ptr,
                        reinterpret_cast<__bf16*>(dqkv.data_ptr()), B, S, NH,
                        static_cast<float>(p), scale,
                        static_cast<uint64_t>(seed),

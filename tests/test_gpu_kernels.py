@@ -213,7 +213,7 @@ def test_attention_forward(B, S, NH):
     H = NH * 64
     qkv = (torch.randn(B, S, 3 * H, device=DEV) * 0.5).bfloat16()
     seqlens = torch.randint(S // 2, S + 1, (B,), device=DEV, dtype=torch.int32)
-    out, lse = ext().attention_fwd(qkv, seqlens, NH, 0.0, 0, 0)
+    out, lse, _mask = ext().attention_fwd(qkv, seqlens, NH, 0.0, 0, 0)
     out_ref = ref.attention(qkv.float(), seqlens, NH, 0.0, False)
     assert rel_err(out, out_ref) < 3e-2
 
@@ -229,7 +229,7 @@ def test_attention_forward_spike():
     qkv[:, S - 3, H : 2 * H] += 40.0
     qkv = qkv.bfloat16()
     seqlens = torch.full((B,), S, device=DEV, dtype=torch.int32)
-    out, _ = ext().attention_fwd(qkv, seqlens, NH, 0.0, 0, 0)
+    out, _, _m = ext().attention_fwd(qkv, seqlens, NH, 0.0, 0, 0)
     out_ref = ref.attention(qkv.float(), seqlens, NH, 0.0, False)
     assert torch.isfinite(out.float()).all()
     assert rel_err(out, out_ref) < 5e-2
@@ -246,9 +246,9 @@ def test_attention_backward(B, S, NH):
     dout = (torch.randn_like(out_ref) * 0.5)
     out_ref.backward(dout)
 
-    out, lse = ext().attention_fwd(qkv, seqlens, NH, 0.0, 0, 0)
+    out, lse, _mask = ext().attention_fwd(qkv, seqlens, NH, 0.0, 0, 0)
     dqkv = ext().attention_bwd(
-        dout.bfloat16(), qkv, seqlens, out, lse, NH, 0.0, 0, 0
+        dout.bfloat16(), qkv, seqlens, out, lse, _mask, NH, 0.0, 0, 0
     )
     assert rel_err(dqkv, qr.grad) < 6e-2
 
@@ -259,24 +259,24 @@ def test_attention_dropout_stats_and_determinism():
     H = NH * 64
     qkv = (torch.randn(B, S, 3 * H, device=DEV) * 0.5).bfloat16()
     seqlens = torch.full((B,), S, device=DEV, dtype=torch.int32)
-    o1, l1 = ext().attention_fwd(qkv, seqlens, NH, p, 99, 1)
-    o2, l2 = ext().attention_fwd(qkv, seqlens, NH, p, 99, 1)
+    o1, l1, _m1 = ext().attention_fwd(qkv, seqlens, NH, p, 99, 1)
+    o2, l2, _m2 = ext().attention_fwd(qkv, seqlens, NH, p, 99, 1)
     assert torch.equal(o1, o2), "same philox state must reproduce"
-    o3, _ = ext().attention_fwd(qkv, seqlens, NH, p, 99, 2)
+    o3, _, _m3 = ext().attention_fwd(qkv, seqlens, NH, p, 99, 2)
     assert not torch.equal(o1, o3), "different offset must differ"
     # dropped-mean check: E[out_p] == out_0 within tolerance
-    o0, _ = ext().attention_fwd(qkv, seqlens, NH, 0.0, 0, 0)
+    o0, _, _m0 = ext().attention_fwd(qkv, seqlens, NH, 0.0, 0, 0)
     acc = torch.zeros_like(o0, dtype=torch.float32)
     n = 32
     for i in range(n):
-        oi, _ = ext().attention_fwd(qkv, seqlens, NH, p, 1234, 100 + i * 10**7)
+        oi, _, _mi = ext().attention_fwd(qkv, seqlens, NH, p, 1234, 100 + i * 10**7)
         acc += oi.float()
     mean = acc / n
     err = (mean - o0.float()).abs().mean() / o0.float().abs().mean()
     assert err < 0.2, f"dropout mean deviates: {err}"
     # backward runs and is finite
     dq = ext().attention_bwd(
-        qkv[..., : H].contiguous(), qkv, seqlens, o1, l1, NH, p, 99, 1
+        qkv[..., : H].contiguous(), qkv, seqlens, o1, l1, _m1, NH, p, 99, 1
     )
     assert torch.isfinite(dq.float()).all()
 
