@@ -234,9 +234,24 @@ class BertLayer(nn.Module):
         self.attention = BertAttention(config)
         self.intermediate = BertIntermediate(config)
         self.output = BertOutput(config)
+        self._ffn_gelu = config.hidden_act in ("gelu", "bias_gelu")
 
     def forward(self, x: torch.Tensor, seqlens: torch.Tensor) -> torch.Tensor:
         attn = self.attention(x, seqlens)
+        if self._ffn_gelu and ops.ffn_supported(attn):
+            # GELU folded into the FFN GEMM epilogues (ops/ffn.py);
+            # FFN2's bias stays fused in the bdrl kernel below
+            y = ops.fused_ffn(
+                attn,
+                self.intermediate.dense_act.weight,
+                self.intermediate.dense_act.bias,
+                self.output.dense.weight,
+            )
+            return ops.fused_bias_dropout_residual_ln(
+                y, self.output.dense.bias, attn,
+                self.output.LayerNorm.weight, self.output.LayerNorm.bias,
+                self.output.dropout_prob, self.training,
+            )
         inter = self.intermediate(attn)
         return self.output(inter, attn)
 

@@ -12,6 +12,12 @@ torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias);
 std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor bias);
 torch::Tensor col_sum(torch::Tensor x);
+std::vector<torch::Tensor> gemm_bias_gelu_fwd(torch::Tensor x,
+                                              torch::Tensor w1,
+                                              torch::Tensor b1);
+std::vector<torch::Tensor> gemm_dgelu_bgrad(torch::Tensor dout,
+                                            torch::Tensor w2,
+                                            torch::Tensor aux);
 std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
     torch::Tensor x, c10::optional<torch::Tensor> bias, torch::Tensor residual,
     torch::Tensor gamma, torch::Tensor beta, double p, double eps,
@@ -80,6 +86,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_gelu_fwd", &bpa::bias_gelu_fwd);
   m.def("bias_gelu_bwd", &bpa::bias_gelu_bwd);
   m.def("col_sum", &bpa::col_sum, "column sum [rows,H] -> fp32 [H]");
+  m.def("gemm_bias_gelu_fwd", &bpa::gemm_bias_gelu_fwd,
+        "hipblaslt GEMM with fused bias+GELU epilogue (+aux)");
+  m.def("gemm_dgelu_bgrad", &bpa::gemm_dgelu_bgrad,
+        "hipblaslt dgrad GEMM with fused dGELU+bias-grad epilogue");
   m.def("bias_dropout_residual_ln_fwd", &bpa::bias_dropout_residual_ln_fwd);
   m.def("bias_dropout_residual_ln_bwd", &bpa::bias_dropout_residual_ln_bwd);
   m.def("embedding_ln_dropout_fwd", &bpa::embedding_ln_dropout_fwd);

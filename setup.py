@@ -22,6 +22,7 @@ sources = [
     "csrc/ops/embedding.hip",
     "csrc/ops/cross_entropy.hip",
     "csrc/ops/attention.hip",
+    "csrc/ops/gemm_epilogue.cpp",
     "csrc/optim/multi_tensor.hip",
     "csrc/tok/tokenizer.cpp",
 ]
@@ -36,6 +37,7 @@ setup(
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17"],
             },
+            libraries=["hipblaslt"],
         )
     ],
     cmdclass={"build_ext": cpp_extension.BuildExtension},
