@@ -51,8 +51,19 @@ class _FusedFFN(torch.autograd.Function):
         return dx, dw1, db1, dw2
 
 
+_aux_supported = None
+
+
 def ffn_supported(x: torch.Tensor) -> bool:
-    return use_native(x)
+    """True only when hipBLASLt offers the GELU_AUX_BIAS/DGELU_BGRAD
+    epilogues (ROCm 7.2's gfx950 library does NOT - probed once; the
+    standalone bias-GELU kernels then run instead)."""
+    global _aux_supported
+    if not use_native(x):
+        return False
+    if _aux_supported is None:
+        _aux_supported = bool(extension().gemm_gelu_aux_supported())
+    return _aux_supported
 
 
 def fused_ffn(

@@ -18,6 +18,7 @@ std::vector<torch::Tensor> gemm_bias_gelu_fwd(torch::Tensor x,
 std::vector<torch::Tensor> gemm_dgelu_bgrad(torch::Tensor dout,
                                             torch::Tensor w2,
                                             torch::Tensor aux);
+bool gemm_gelu_aux_supported();
 std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
     torch::Tensor x, c10::optional<torch::Tensor> bias, torch::Tensor residual,
     torch::Tensor gamma, torch::Tensor beta, double p, double eps,
@@ -90,6 +91,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "hipblaslt GEMM with fused bias+GELU epilogue (+aux)");
   m.def("gemm_dgelu_bgrad", &bpa::gemm_dgelu_bgrad,
         "hipblaslt dgrad GEMM with fused dGELU+bias-grad epilogue");
+  m.def("gemm_gelu_aux_supported", &bpa::gemm_gelu_aux_supported);
   m.def("bias_dropout_residual_ln_fwd", &bpa::bias_dropout_residual_ln_fwd);
   m.def("bias_dropout_residual_ln_bwd", &bpa::bias_dropout_residual_ln_bwd);
   m.def("embedding_ln_dropout_fwd", &bpa::embedding_ln_dropout_fwd);

@@ -247,4 +247,56 @@ std::vector<torch::Tensor> gemm_dgelu_bgrad(torch::Tensor dout,
   return {dpre, db1};
 }
 
+// Runtime capability probe: does this hipBLASLt build offer algos for
+// the training-side GELU fusion (GELU_AUX_BIAS + DGELU_BGRAD) on these
+// shapes? ROCm 7.2's gfx950 library answers NO (plain GELU_BIAS only,
+// which cannot serve training because backward needs the
+// pre-activation); the fused-FFN path auto-disables via this check and
+// the standalone bias-GELU kernels run instead.
+bool gemm_gelu_aux_supported() {
+  static int cached = -1;
+  if (cached >= 0) return cached != 0;
+  hipblasLtMatmulDesc_t desc;
+  if (hipblasLtMatmulDescCreate(&desc, HIPBLAS_COMPUTE_32F, HIP_R_32F) !=
+      HIPBLAS_STATUS_SUCCESS) {
+    cached = 0;
+    return false;
+  }
+  int32_t ta = HIPBLAS_OP_T, tb = HIPBLAS_OP_N;
+  hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_TRANSA, &ta, 4);
+  hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_TRANSB, &tb, 4);
+  uint32_t epi = HIPBLASLT_EPILOGUE_GELU_AUX_BIAS;
+  hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi,
+                                  4);
+  void* dummy = reinterpret_cast<void*>(0x1000);
+  hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER,
+                                  &dummy, sizeof(dummy));
+  hipblasLtMatmulDescSetAttribute(
+      desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_POINTER, &dummy,
+      sizeof(dummy));
+  int64_t ld = 4096;
+  hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_LD,
+                                  &ld, 8);
+  hipblasLtMatrixLayout_t la, lb, lc;
+  hipblasLtMatrixLayoutCreate(&la, HIP_R_16BF, 1024, 4096, 1024);
+  hipblasLtMatrixLayoutCreate(&lb, HIP_R_16BF, 1024, 4096, 1024);
+  hipblasLtMatrixLayoutCreate(&lc, HIP_R_16BF, 4096, 4096, 4096);
+  hipblasLtMatmulPreference_t pref;
+  hipblasLtMatmulPreferenceCreate(&pref);
+  size_t ws = kWorkspaceBytes;
+  hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws));
+  hipblasLtMatmulHeuristicResult_t res[4];
+  int found = 0;
+  hipblasLtMatmulAlgoGetHeuristic(handle(), desc, la, lb, lc, lc, pref, 4,
+                                  res, &found);
+  hipblasLtMatmulPreferenceDestroy(pref);
+  hipblasLtMatrixLayoutDestroy(la);
+  hipblasLtMatrixLayoutDestroy(lb);
+  hipblasLtMatrixLayoutDestroy(lc);
+  hipblasLtMatmulDescDestroy(desc);
+  cached = found > 0 ? 1 : 0;
+  return cached != 0;
+}
+
 }  // namespace bpa

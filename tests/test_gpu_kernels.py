@@ -559,6 +559,10 @@ def test_kfac_gpu_step():
 # ---------------------------------------------------------------------------
 def test_fused_ffn_matches_eager():
     from bert_pytorch_amd.ops import fused_ffn
+    from bert_pytorch_amd.ops.ffn import ffn_supported
+
+    if not ffn_supported(torch.zeros(1, device=DEV)):
+        pytest.skip("hipBLASLt build lacks GELU_AUX_BIAS/DGELU_BGRAD")
 
     torch.manual_seed(11)
     M, H, F = 512, 1024, 4096
@@ -588,8 +592,8 @@ def test_fused_ffn_matches_eager():
 
 
 def test_model_uses_fused_ffn_path():
-    """BertLayer takes the fused-FFN branch on device and still matches
-    the CPU forward within bf16 tolerance (gelu-flavor included)."""
+    """BertLayer forward on device matches the CPU forward (takes the
+    fused-FFN branch when the hipBLASLt build supports it)."""
     from bert_pytorch_amd.config import BertConfig
     from bert_pytorch_amd.models import BertForPreTraining
 
