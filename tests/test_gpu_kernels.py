@@ -207,7 +207,7 @@ def test_cross_entropy_all_ignored():
 # ---------------------------------------------------------------------------
 # attention
 # ---------------------------------------------------------------------------
-@pytest.mark.parametrize("B,S,NH", [(4, 128, 16), (2, 512, 16), (3, 128, 4)])
+@pytest.mark.parametrize("B,S,NH", [(4, 128, 16), (2, 512, 16), (3, 128, 4), (3, 80, 4)])
 def test_attention_forward(B, S, NH):
     torch.manual_seed(7)
     H = NH * 64
@@ -235,7 +235,7 @@ def test_attention_forward_spike():
     assert rel_err(out, out_ref) < 5e-2
 
 
-@pytest.mark.parametrize("B,S,NH", [(4, 128, 8), (2, 512, 4)])
+@pytest.mark.parametrize("B,S,NH", [(4, 128, 8), (2, 512, 4), (3, 80, 4)])
 def test_attention_backward(B, S, NH):
     torch.manual_seed(9)
     H = NH * 64
