@@ -59,6 +59,27 @@ __device__ __forceinline__ bf16x8 frag_row(const __bf16* row, int base,
   return r.v;
 }
 
+// Dropout keep-mask generation: one Philox call per 4 bytes. The
+// counter for quad j is offset + j, which matches the per-(bh,q,key)
+// counters the consumers index with (drop_base(bh) + q*(S/4) + key/4
+// is exactly the linear quad index). Running this as its own
+// elementwise kernel keeps the MFMA kernels free of RNG VALU work.
+__global__ void dropout_mask_kernel(uint8_t* __restrict__ mask,
+                                    int64_t quads, float p, uint64_t seed,
+                                    uint64_t offset) {
+  const int64_t i =
+      static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  if (i >= quads) return;
+  Philox philox(seed);
+  uint32_t r4[4];
+  philox(offset + i, r4);
+  uint8_t mb[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) mb[j] = u32_to_uniform(r4[j]) >= p ? 1 : 0;
+  *reinterpret_cast<uint32_t*>(mask + i * 4) =
+      *reinterpret_cast<const uint32_t*>(mb);
+}
+
 // ---------------------------------------------------------------------------
 // forward
 // ---------------------------------------------------------------------------
@@ -66,8 +87,8 @@ template <bool TRAIN_DROP>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const __bf16* __restrict__ qkv, const int* __restrict__ seqlens,
     __bf16* __restrict__ out, float* __restrict__ lse_out,
-    uint8_t* __restrict__ dmask, int B, int S, int NH, float p, float scale,
-    uint64_t seed, uint64_t offset) {
+    const uint8_t* __restrict__ dmask, int B, int S, int NH, float p,
+    float scale, uint64_t seed, uint64_t offset) {
   // 4 waves x 32 q-rows (two 16-row subtiles per wave): the K/V tile is
   // staged once per 128 q-rows, each K/V fragment read feeds TWO
   // independent MFMA chains, and the softmax work of the two subtiles
@@ -85,11 +106,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const __bf16* vbase = qbase + 2 * H;
   const int slen = seqlens[b];
   const float inv_keep = TRAIN_DROP ? 1.f / (1.f - p) : 1.f;
-  Philox philox(seed);
-  const int s4 = S >> 2;
-  const uint64_t drop_base =
-      offset + static_cast<uint64_t>(bh) * S * s4;
-  uint8_t* mask_base =
+  const uint8_t* mask_base =
       TRAIN_DROP ? dmask + static_cast<int64_t>(bh) * S * S : nullptr;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -203,10 +220,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
           acc_o[sub][n][r] *= alpha_lds[(wave * 2 + sub) * 16 + g * 4 + r];
 
     if (TRAIN_DROP) {
-      // one Philox call per key-quad (all 4 words used by THIS lane);
-      // the keep-mask is stored as bytes so the backward kernels read
-      // it instead of regenerating (dropout RNG was ~2x fwd cost and
-      // similar in bwd/dq)
+      // keep-mask pre-generated by dropout_mask_kernel (no RNG VALU
+      // work in the MFMA kernels); one 4-byte read per key-quad
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {
         const int q_abs = q_row[sub];
@@ -214,21 +229,15 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
         for (int t = 0; t < 4; ++t) {
           const int key0 = k0 + t * 16 + g * 4;
-          uint32_t r4[4];
-          philox(drop_base + static_cast<uint64_t>(q_abs) * s4 + (key0 >> 2),
-                 r4);
-          uint8_t mb[4];
+          uint8_t mb[4] = {1, 1, 1, 1};
+          if (q_ok && key0 + 4 <= S) {  // quad fully inside the row
+            *reinterpret_cast<uint32_t*>(mb) =
+                *reinterpret_cast<const uint32_t*>(
+                    mask_base + static_cast<int64_t>(q_abs) * S + key0);
+          }
 #pragma unroll
-          for (int j = 0; j < 4; ++j) {
-            const bool keep = u32_to_uniform(r4[j]) >= p;
-            mb[j] = keep ? 1 : 0;
-            sv[sub][t][j] = keep ? sv[sub][t][j] * inv_keep : 0.f;
-          }
-          if (q_ok) {
-            *reinterpret_cast<uint32_t*>(
-                mask_base + static_cast<int64_t>(q_abs) * S + key0) =
-                *reinterpret_cast<const uint32_t*>(mb);
-          }
+          for (int j = 0; j < 4; ++j)
+            sv[sub][t][j] = mb[j] ? sv[sub][t][j] * inv_keep : 0.f;
         }
       }
     }
@@ -446,7 +455,7 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
       for (int sub = 0; sub < 2; ++sub) {
         // read the forward's stored keep-mask: one 4-byte load per quad
         uint8_t mb[4] = {1, 1, 1, 1};
-        if (TRAIN_DROP && q_row[sub] < S) {
+        if (TRAIN_DROP && q_row[sub] < S && k0 + t * 16 + g * 4 + 4 <= S) {
           *reinterpret_cast<uint32_t*>(mb) = *reinterpret_cast<const uint32_t*>(
               mask_base + static_cast<int64_t>(q_row[sub]) * S + k0 + t * 16 +
               g * 4);
@@ -738,6 +747,12 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor qkv,
   const size_t lds =
       2 * 64 * kStride * sizeof(__bf16) + 4 * 64 * sizeof(float);
   if (train_drop) {
+    const int64_t quads = static_cast<int64_t>(B) * NH * S * S / 4;
+    hipLaunchKernelGGL(dropout_mask_kernel,
+                       dim3((quads + 255) / 256), dim3(256), 0, stream,
+                       dmask.data_ptr<uint8_t>(), quads,
+                       static_cast<float>(p), static_cast<uint64_t>(seed),
+                       static_cast<uint64_t>(offset));
     hipLaunchKernelGGL((attn_fwd_kernel<true>), grid, block, lds, stream,
                        reinterpret_cast<const __bf16*>(qkv.data_ptr()),
                        seql.data_ptr<int>(),
