@@ -23,7 +23,7 @@ from __future__ import annotations
 
 import struct
 import zlib
-from typing import Dict, Optional
+from typing import Dict
 
 import numpy as np
 

@@ -18,7 +18,7 @@ import json
 import math
 import re
 import string
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
 from .tokenization import BasicTokenizer

@@ -8,7 +8,6 @@ inside the fused update. Kernel source: csrc/optim/multi_tensor.hip.
 
 from __future__ import annotations
 
-import os
 from typing import List
 
 import torch
