@@ -8,8 +8,11 @@
 // wave. K and V^T staged in LDS (+8 bf16 row pad, conflict-free b64
 // reads); online softmax with the swapped-QK^T trick (S^T = mfma(K, Q))
 // so each lane's P scores chain directly into the PV A-fragment; the
-// padding mask enters as per-sequence valid lengths; dropout by
-// Philox4x32-10 regenerated (not stored) in backward.
+// padding mask enters as per-sequence valid lengths. Dropout keep-masks
+// are pre-generated once per forward by dropout_mask_kernel (Philox
+// 4x32-10, one byte per score) and the [B*NH, S, S] byte tensor is read
+// by forward and both backward kernels — measurably cheaper than
+// regenerating the RNG stream in each of the three consumers.
 //
 // Backward (flash-2 style, atomic-free): two MFMA kernels.
 // * attn_bwd_kernel: grid (S/64 kv-tiles, B*heads); each block owns one
