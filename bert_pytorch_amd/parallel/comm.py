@@ -111,7 +111,10 @@ def wrap_ddp(
     """
     if get_world_size() == 1:
         return model
-    device_ids = [local_rank] if torch.cuda.is_available() else None
+    # Key off the module's actual device, not cuda.is_available(): a
+    # CPU/gloo module on a GPU box must not get device_ids.
+    on_cuda = any(p.is_cuda for p in model.parameters())
+    device_ids = [local_rank] if on_cuda else None
     ddp = torch.nn.parallel.DistributedDataParallel(
         model,
         device_ids=device_ids,
