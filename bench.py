@@ -107,7 +107,13 @@ def main():
     if accum <= 0:
         accum = max(1, -(-phase["named_global"] // (world * bsz)))
 
-    config = BertConfig.from_json_file(args.model_config)
+    cfg_path = args.model_config
+    if not os.path.exists(cfg_path):
+        # resolve relative to this file so bench.py works from any cwd
+        here = os.path.join(os.path.dirname(os.path.abspath(__file__)), cfg_path)
+        if os.path.exists(here):
+            cfg_path = here
+    config = BertConfig.from_json_file(cfg_path)
     if config.vocab_size % 64:
         config.vocab_size += 64 - config.vocab_size % 64
     model = BertForPreTraining(config).to(device)
