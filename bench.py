@@ -74,14 +74,15 @@ def parse_args():
 
 
 def make_batch(gen, device, bsz, seq, vocab, max_pred):
-    ids = torch.randint(1000, vocab, (bsz, seq), generator=gen, device=device)
+    lo = min(1000, vocab - 1)
+    ids = torch.randint(lo, vocab, (bsz, seq), generator=gen, device=device)
     ids[:, 0] = 101
     tt = torch.zeros_like(ids)
     tt[:, seq // 2 :] = 1
     mask = torch.ones_like(ids)
     labels = torch.full((bsz, seq), -1, dtype=torch.long, device=device)
     pos = torch.rand(bsz, seq, generator=gen, device=device).argsort(dim=1)[:, :max_pred]
-    vals = torch.randint(1000, vocab, (bsz, max_pred), generator=gen, device=device)
+    vals = torch.randint(lo, vocab, (bsz, max_pred), generator=gen, device=device)
     labels.scatter_(1, pos, vals)
     ids.scatter_(1, pos, torch.full_like(vals, 103))
     nsp = torch.randint(0, 2, (bsz,), generator=gen, device=device)
@@ -136,7 +137,8 @@ def main():
     )
     scheduler = PolyWarmUpScheduler(optimizer, warmup=0.2843, total_steps=7038)
     gen = torch.Generator(device=device).manual_seed(args.seed + rank)
-    vocab_unpadded = 30522
+    # BERT uncased real vocab (30522) unless the model config is smaller
+    vocab_unpadded = min(30522, config.vocab_size)
 
     ac_enabled = use_cuda and not args.pure_bf16
 
