@@ -169,7 +169,9 @@ def main(args=None):
                 global_step += 1
             f1 = evaluate(model, eval_loader, device, mixed, train_set.num_labels)
             log.log("eval", epoch, macro_f1=f1, loss=float(loss))
-        log.info("e2e_train_time=%.1fs", time.perf_counter() - start)
+        train_time = time.perf_counter() - start
+        log.info("e2e_train_time=%.1fs training_sequences_per_second=%.1f",
+                 train_time, args.epochs * len(train_set) / train_time)
         if comm.is_main_process():
             torch.save({"model": model.state_dict()},
                        os.path.join(args.output_dir, "pytorch_model.bin"))
