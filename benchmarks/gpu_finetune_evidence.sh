@@ -11,7 +11,7 @@ OUT=${OUT:-/tmp/bpa_finetune}
 LOGS=${LOGS:-gpurun_out}
 mkdir -p "$LOGS"
 python benchmarks/gen_finetune_synth.py --out "$OUT" \
-    --squad_train 2048 --squad_predict 1024 --ner_train 4000 --ner_valid 800
+    --squad_train 8192 --squad_predict 4096 --ner_train 12000 --ner_valid 800
 
 timeout 900 python run_squad.py \
     --model_config_file config/bert_large_uncased_config.json \
@@ -19,7 +19,7 @@ timeout 900 python run_squad.py \
     --train_file "$OUT/squad_train.json" \
     --predict_file "$OUT/squad_predict.json" \
     --do_train --do_predict --do_eval --bf16 \
-    --max_steps 40 --train_batch_size 32 --predict_batch_size 32 \
+    --max_steps 250 --train_batch_size 32 --predict_batch_size 32 \
     --max_seq_length 384 --doc_stride 128 \
     --output_dir "$OUT/sq_out" > "$LOGS/squad_bertlarge_gpu.log" 2>&1
 echo "SQUAD_RC=$?"
