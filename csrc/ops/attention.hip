@@ -642,6 +642,25 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
         lse_lds[tid] = (qr < S) ? lse[static_cast<int64_t>(bh) * S + qr] : 0.f;
         dlt_lds[tid] = (qr < S) ? delta[static_cast<int64_t>(bh) * S + qr] : 0.f;
       }
+      if (TRAIN_DROP) {
+        // stage the keep-mask tile [64 q][128 keys]: 256 threads x two
+        // 16-byte chunks; chunks past the row end fill with ones (those
+        // keys are already zeroed by the kvalid/slen guards)
+        const int mrow = tid >> 2, mc = (tid & 3) * 32;
+        const int qrow_m = q0 + mrow;
+        uint4 mv0 = {0x01010101u, 0x01010101u, 0x01010101u, 0x01010101u};
+        uint4 mv1 = mv0;
+        if (qrow_m < S) {
+          const uint8_t* src =
+              mask_base + static_cast<int64_t>(qrow_m) * S + k0 + mc;
+          if (k0 + mc + 16 <= S)
+            mv0 = *reinterpret_cast<const uint4*>(src);
+          if (k0 + mc + 32 <= S)
+            mv1 = *reinterpret_cast<const uint4*>(src + 16);
+        }
+        *reinterpret_cast<uint4*>(&mk_lds[mrow * 128 + mc]) = mv0;
+        *reinterpret_cast<uint4*>(&mk_lds[mrow * 128 + mc + 16]) = mv1;
+      }
     }
     __syncthreads();
 

@@ -613,3 +613,46 @@ def test_model_uses_fused_ffn_path():
     with torch.no_grad():
         scores_cpu, _ = cpu_model(ids.cpu(), None, mask.cpu())
     assert rel_err(scores.cpu(), scores_cpu) < 5e-2
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("B,S,NH", [(2, 128, 4), (1, 512, 2)])
+def test_attention_dropout_numerics(B, S, NH):
+    """Forward AND backward with dropout must match an eager fp32
+    reference conditioned on the kernel's own stored keep-mask (this is
+    the regression test for the bwd mask-staging bug, where attn_bwd
+    read an unwritten LDS tile when p > 0)."""
+    torch.manual_seed(21)
+    H = NH * 64
+    p = 0.1
+    p_q = round(p * 256) / 256  # the kernel's 8-bit-quantized drop prob
+    qkv = (torch.randn(B, S, 3 * H, device=DEV) * 0.5).bfloat16()
+    seqlens = torch.randint(S // 2, S + 1, (B,), device=DEV, dtype=torch.int32)
+
+    out, lse, dmask = ext().attention_fwd(qkv, seqlens, NH, p, 77, 5)
+    keep = dmask.view(B, NH, S, S).float()
+    # keep-rate sanity: mean within 3 sigma of 1 - p_q
+    n = keep.numel()
+    assert abs(keep.mean().item() - (1 - p_q)) < 4 * (p_q * (1 - p_q) / n) ** 0.5 + 1e-3
+
+    qr = qkv.float().detach().requires_grad_(True)
+    q, k, v = qr.split(H, dim=-1)
+
+    def shape(t):
+        return t.view(B, S, NH, 64).transpose(1, 2)
+
+    scores = torch.matmul(shape(q), shape(k).transpose(-1, -2)) / 8.0
+    key_pad = torch.arange(S, device=DEV).unsqueeze(0) >= seqlens.unsqueeze(1)
+    scores = scores.masked_fill(key_pad[:, None, None, :], -10000.0)
+    probs = torch.softmax(scores, dim=-1) * keep / (1 - p_q)
+    out_ref = (
+        torch.matmul(probs, shape(v)).transpose(1, 2).reshape(B, S, H)
+    )
+    assert rel_err(out, out_ref) < 3e-2
+
+    dout = torch.randn_like(out_ref) * 0.5
+    out_ref.backward(dout)
+    dqkv = ext().attention_bwd(
+        dout.bfloat16(), qkv, seqlens, out, lse, dmask, NH, p, 77, 5
+    )
+    assert rel_err(dqkv, qr.grad) < 6e-2
