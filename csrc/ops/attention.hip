@@ -10,9 +10,11 @@
 // so each lane's P scores chain directly into the PV A-fragment; the
 // padding mask enters as per-sequence valid lengths. Dropout keep-masks
 // are pre-generated once per forward by dropout_mask_kernel (Philox
-// 4x32-10, one byte per score) and the [B*NH, S, S] byte tensor is read
-// by forward and both backward kernels — measurably cheaper than
-// regenerating the RNG stream in each of the three consumers.
+// 4x32-10, one BIT per score, rows padded to whole 32-bit words) and
+// the [B*NH, S, ceil(S/32)] word tensor is read by forward and both
+// backward kernels — measurably cheaper than regenerating the RNG
+// stream in each of the three consumers, and 8x less mask traffic
+// than the byte-mask variant.
 //
 // Backward (flash-2 style, atomic-free): two MFMA kernels.
 // * attn_bwd_kernel: grid (S/64 kv-tiles, B*heads); each block owns one
@@ -65,35 +67,38 @@ __device__ __forceinline__ bf16x8 frag_row(const __bf16* row, int base,
   return r.v;
 }
 
-// Dropout keep-mask generation: one Philox call per SIXTEEN bytes —
-// each of the four 32-bit outputs yields four keep decisions by
+// Dropout keep-mask generation, BIT-packed: one keep decision per BIT
+// (LSB-first within each uint32 word), one thread per word. Each of
+// the eight 32-bit Philox outputs (two calls) yields four decisions by
 // comparing its bytes against an 8-bit threshold t = round(p*256).
 // The host quantizes p to t/256 first (quantize_drop_p) and every
 // consumer's 1/(1-p) uses the SAME quantized p, so dropout stays
 // exactly unbiased; the ~2^-8 quantization of the drop probability
-// itself (0.1 -> 0.1016) is training-irrelevant. The previous
-// one-Philox-per-4-bytes version was VALU-bound at ~1.4 TB/s (profile:
-// 2.9% of phase-2 GPU time); 4x fewer Philox calls + uint4 stores fix
-// that. Counter for 16-byte group i is offset + i, deterministic in
-// (seed, offset). Running this as its own elementwise kernel keeps the
-// MFMA kernels free of RNG VALU work.
-__global__ void dropout_mask_kernel(uint8_t* __restrict__ mask,
-                                    int64_t groups, uint32_t thresh,
+// itself (0.1 -> 0.1016) is training-irrelevant. History: the original
+// one-Philox-per-4-BYTES version was VALU-bound at ~1.4 TB/s and 2.9%
+// of phase-2 GPU time; bytes->bits also cuts the mask footprint and
+// the three consumers' read traffic 8x (24 layers x [B*NH,S,S] masks
+// alive per micro-batch = 1.6 GB at phase 2 as bytes, 200 MB as bits).
+// Rows are padded to whole words (stride Sw = ceil(S/32)) so no quad
+// ever straddles a word. Counter for word i is offset + 2i (+1),
+// deterministic in (seed, offset). Running this as its own elementwise
+// kernel keeps the MFMA kernels free of RNG VALU work.
+__global__ void dropout_mask_kernel(uint32_t* __restrict__ mask,
+                                    int64_t words, uint32_t thresh,
                                     uint64_t seed, uint64_t offset) {
   const int64_t i =
       static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
-  if (i >= groups) return;
+  if (i >= words) return;
   Philox philox(seed);
-  uint32_t r4[4];
-  philox(offset + i, r4);
-  uint8_t mb[16];
+  uint32_t r[8];
+  philox(offset + 2 * i, r);
+  philox(offset + 2 * i + 1, r + 4);
+  uint32_t bits = 0;
 #pragma unroll
-  for (int w = 0; w < 4; ++w)
-#pragma unroll
-    for (int j = 0; j < 4; ++j)
-      mb[w * 4 + j] = ((r4[w] >> (8 * j)) & 0xFFu) >= thresh ? 1 : 0;
-  *reinterpret_cast<uint4*>(mask + i * 16) =
-      *reinterpret_cast<const uint4*>(mb);
+  for (int j = 0; j < 32; ++j)
+    bits |= (((r[j >> 2] >> (8 * (j & 3))) & 0xFFu) >= thresh ? 1u : 0u)
+            << j;
+  mask[i] = bits;
 }
 
 // Quantize a dropout probability to the 8-bit threshold grid the mask
@@ -112,7 +117,7 @@ template <bool TRAIN_DROP>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const __bf16* __restrict__ qkv, const int* __restrict__ seqlens,
     __bf16* __restrict__ out, float* __restrict__ lse_out,
-    const uint8_t* __restrict__ dmask, int B, int S, int NH, float p,
+    const uint32_t* __restrict__ dmask, int B, int S, int NH, float p,
     float scale, uint64_t seed, uint64_t offset) {
   // 4 waves x 32 q-rows (two 16-row subtiles per wave): the K/V tile is
   // staged once per 128 q-rows, each K/V fragment read feeds TWO
@@ -131,8 +136,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const __bf16* vbase = qbase + 2 * H;
   const int slen = seqlens[b];
   const float inv_keep = TRAIN_DROP ? 1.f / (1.f - p) : 1.f;
-  const uint8_t* mask_base =
-      TRAIN_DROP ? dmask + static_cast<int64_t>(bh) * S * S : nullptr;
+  const int Sw = (S + 31) >> 5;  // mask row stride in words
+  const uint32_t* mask_base =
+      TRAIN_DROP ? dmask + static_cast<int64_t>(bh) * S * Sw : nullptr;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* K_lds = reinterpret_cast<__bf16*>(smem);             // [64][72]
@@ -254,15 +260,15 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
         for (int t = 0; t < 4; ++t) {
           const int key0 = k0 + t * 16 + g * 4;
-          uint8_t mb[4] = {1, 1, 1, 1};
-          if (q_ok && key0 + 4 <= S) {  // quad fully inside the row
-            *reinterpret_cast<uint32_t*>(mb) =
-                *reinterpret_cast<const uint32_t*>(
-                    mask_base + static_cast<int64_t>(q_abs) * S + key0);
-          }
+          uint32_t nib = 0xF;  // keys past the row end: keep (masked anyway)
+          if (q_ok && key0 < S)  // quads are word-aligned, never straddle
+            nib = (mask_base[static_cast<int64_t>(q_abs) * Sw +
+                             (key0 >> 5)] >>
+                   (key0 & 31)) &
+                  0xF;
 #pragma unroll
           for (int j = 0; j < 4; ++j)
-            sv[sub][t][j] = mb[j] ? sv[sub][t][j] * inv_keep : 0.f;
+            sv[sub][t][j] = (nib >> j) & 1 ? sv[sub][t][j] * inv_keep : 0.f;
         }
       }
     }
@@ -370,7 +376,7 @@ template <bool TRAIN_DROP>
 __global__ __launch_bounds__(256) void attn_dq_kernel(
     const __bf16* __restrict__ dout, const __bf16* __restrict__ qkv,
     const int* __restrict__ seqlens, const float* __restrict__ lse,
-    const float* __restrict__ delta, const uint8_t* __restrict__ dmask,
+    const float* __restrict__ delta, const uint32_t* __restrict__ dmask,
     __bf16* __restrict__ dqkv, int B, int S, int NH, float p, float scale,
     uint64_t seed, uint64_t offset) {
   // 4 waves x 32 q-rows (two 16-row subtiles per wave), mirroring
@@ -390,8 +396,9 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
   const __bf16* dobase = dout + static_cast<int64_t>(b) * S * H + h * 64;
   const int slen = seqlens[b];
   const float inv_keep = TRAIN_DROP ? 1.f / (1.f - p) : 1.f;
-  const uint8_t* mask_base =
-      TRAIN_DROP ? dmask + static_cast<int64_t>(bh) * S * S : nullptr;
+  const int Sw = (S + 31) >> 5;  // mask row stride in words
+  const uint32_t* mask_base =
+      TRAIN_DROP ? dmask + static_cast<int64_t>(bh) * S * Sw : nullptr;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* K_lds = reinterpret_cast<__bf16*>(smem);   // [64][72] natural
@@ -478,12 +485,14 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
       }
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {
-        // read the forward's stored keep-mask: one 4-byte load per quad
-        uint8_t mb[4] = {1, 1, 1, 1};
-        if (TRAIN_DROP && q_row[sub] < S && k0 + t * 16 + g * 4 + 4 <= S) {
-          *reinterpret_cast<uint32_t*>(mb) = *reinterpret_cast<const uint32_t*>(
-              mask_base + static_cast<int64_t>(q_row[sub]) * S + k0 + t * 16 +
-              g * 4);
+        // read the forward's stored keep-mask: one word load per quad
+        uint32_t nib = 0xF;
+        if (TRAIN_DROP && q_row[sub] < S && k0 + t * 16 + g * 4 < S) {
+          const int key0 = k0 + t * 16 + g * 4;
+          nib = (mask_base[static_cast<int64_t>(q_row[sub]) * Sw +
+                           (key0 >> 5)] >>
+                 (key0 & 31)) &
+                0xF;
         }
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
@@ -493,7 +502,7 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
               valid ? __expf(sacc[sub][r] * scale - lse_q[sub]) : 0.f;
           float dpd = dpacc[sub][r];
           if (TRAIN_DROP) {
-            dpd = mb[r] ? dpd * inv_keep : 0.f;
+            dpd = (nib >> r) & 1 ? dpd * inv_keep : 0.f;
           }
           dsfrag[sub][t >> 1].e[(t & 1) * 4 + r] =
               __bf16(pr * (dpd - dlt_q[sub]) * scale);
@@ -533,7 +542,7 @@ template <bool TRAIN_DROP>
 __global__ __launch_bounds__(256) void attn_bwd_kernel(
     const __bf16* __restrict__ dout, const __bf16* __restrict__ qkv,
     const int* __restrict__ seqlens, const float* __restrict__ lse,
-    const float* __restrict__ delta, const uint8_t* __restrict__ dmask,
+    const float* __restrict__ delta, const uint32_t* __restrict__ dmask,
     __bf16* __restrict__ dqkv, int B, int S, int NH, float p, float scale,
     uint64_t seed, uint64_t offset) {
   // 4 waves x 32 keys (two 16-key subtiles per wave): the heavy per-
@@ -553,8 +562,9 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   const __bf16* dobase = dout + static_cast<int64_t>(b) * S * H + h * 64;
   const int slen = seqlens[b];
   const float inv_keep = TRAIN_DROP ? 1.f / (1.f - p) : 1.f;
-  const uint8_t* mask_base =
-      TRAIN_DROP ? dmask + static_cast<int64_t>(bh) * S * S : nullptr;
+  const int Sw = (S + 31) >> 5;  // mask row stride in words
+  const uint32_t* mask_base =
+      TRAIN_DROP ? dmask + static_cast<int64_t>(bh) * S * Sw : nullptr;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* K_lds = reinterpret_cast<__bf16*>(smem);   // [128][72] natural
@@ -565,9 +575,10 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   __bf16* dOt_lds = dO_lds + 64 * kStride;           // [64][72] transposed
   float* lse_lds = reinterpret_cast<float*>(dOt_lds + 64 * kStride);  // [64]
   float* dlt_lds = lse_lds + 64;                                      // [64]
-  // keep-mask tile [64 q][128 keys] staged per q-tile (the raw global
-  // reads are byte columns - latency-bound at this kernel's occupancy)
-  uint8_t* mk_lds = reinterpret_cast<uint8_t*>(dlt_lds + 64);
+  // keep-mask tile [64 q][128 keys] as bits, [64][4] words, staged per
+  // q-tile (raw global reads were byte columns - latency-bound at this
+  // kernel's occupancy; 16 lanes share each word via LDS broadcast)
+  uint32_t* mk_lds = reinterpret_cast<uint32_t*>(dlt_lds + 64);
 
   // stage K and V (natural) once: two 64-row passes
 #pragma unroll
@@ -643,23 +654,16 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
         dlt_lds[tid] = (qr < S) ? delta[static_cast<int64_t>(bh) * S + qr] : 0.f;
       }
       if (TRAIN_DROP) {
-        // stage the keep-mask tile [64 q][128 keys]: 256 threads x two
-        // 16-byte chunks; chunks past the row end fill with ones (those
-        // keys are already zeroed by the kvalid/slen guards)
-        const int mrow = tid >> 2, mc = (tid & 3) * 32;
+        // stage the keep-mask tile [64 q][128 keys] = [64][4] words:
+        // one word per thread; words past the row end fill with ones
+        // (those keys are already zeroed by the kvalid/slen guards)
+        const int mrow = tid >> 2, mw = tid & 3;
         const int qrow_m = q0 + mrow;
-        uint4 mv0 = {0x01010101u, 0x01010101u, 0x01010101u, 0x01010101u};
-        uint4 mv1 = mv0;
-        if (qrow_m < S) {
-          const uint8_t* src =
-              mask_base + static_cast<int64_t>(qrow_m) * S + k0 + mc;
-          if (k0 + mc + 16 <= S)
-            mv0 = *reinterpret_cast<const uint4*>(src);
-          if (k0 + mc + 32 <= S)
-            mv1 = *reinterpret_cast<const uint4*>(src + 16);
-        }
-        *reinterpret_cast<uint4*>(&mk_lds[mrow * 128 + mc]) = mv0;
-        *reinterpret_cast<uint4*>(&mk_lds[mrow * 128 + mc + 16]) = mv1;
+        const int kw = (k0 >> 5) + mw;  // k0 is a multiple of 128
+        uint32_t bits = 0xFFFFFFFFu;
+        if (qrow_m < S && kw < Sw)
+          bits = mask_base[static_cast<int64_t>(qrow_m) * Sw + kw];
+        mk_lds[mrow * 4 + mw] = bits;
       }
     }
     __syncthreads();
@@ -710,9 +714,11 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
           float dpd = dp[sub][r];
           float pkeep = pr;
           if (TRAIN_DROP) {
+            const int key_l = sub ? key_local1 : key_local0;
             const bool keep =
-                mk_lds[(mq * 16 + g * 4 + r) * 128 +
-                       (sub ? key_local1 : key_local0)] != 0;
+                (mk_lds[(mq * 16 + g * 4 + r) * 4 + (key_l >> 5)] >>
+                 (key_l & 31)) &
+                1;
             dpd = keep ? dpd * inv_keep : 0.f;
             pkeep = keep ? pr * inv_keep : 0.f;  // dropped P feeds dV
           }
@@ -779,32 +785,34 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor qkv,
   auto out = torch::empty({B, S, H}, qkv.options());
   auto lse = torch::empty({B * NH, S}, qkv.options().dtype(torch::kFloat32));
   const bool train_drop = p > 0.0;
-  // keep-mask bytes [B*NH, S, S]: written once here, read (not
-  // regenerated) by both backward kernels
+  // keep-mask BITS [B*NH, S, ceil(S/32)] int32 words: written once
+  // here, read (not regenerated) by both backward kernels
+  const int Sw = (S + 31) / 32;
   auto dmask = train_drop
-                   ? torch::empty({static_cast<int64_t>(B) * NH, S, S},
-                                  qkv.options().dtype(torch::kUInt8))
-                   : torch::empty({0}, qkv.options().dtype(torch::kUInt8));
+                   ? torch::empty({static_cast<int64_t>(B) * NH, S, Sw},
+                                  qkv.options().dtype(torch::kInt32))
+                   : torch::empty({0}, qkv.options().dtype(torch::kInt32));
   const float scale = 1.0f / sqrtf(64.f);
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid((S + 127) / 128, B * NH), block(256);  // 128 q-rows per block
   const size_t lds =
       2 * 64 * kStride * sizeof(__bf16) + 4 * 64 * sizeof(float);
   if (train_drop) {
-    // S%16==0 so S*S (and groups of 16 bytes) divide evenly
     const auto [thresh, p_q] = quantize_drop_p(p);
-    const int64_t groups = static_cast<int64_t>(B) * NH * S * S / 16;
+    const int64_t words = static_cast<int64_t>(B) * NH * S * Sw;
     hipLaunchKernelGGL(dropout_mask_kernel,
-                       dim3((groups + 255) / 256), dim3(256), 0, stream,
-                       dmask.data_ptr<uint8_t>(), groups, thresh,
-                       static_cast<uint64_t>(seed),
+                       dim3((words + 255) / 256), dim3(256), 0, stream,
+                       reinterpret_cast<uint32_t*>(dmask.data_ptr<int>()),
+                       words, thresh, static_cast<uint64_t>(seed),
                        static_cast<uint64_t>(offset));
     hipLaunchKernelGGL((attn_fwd_kernel<true>), grid, block, lds, stream,
                        reinterpret_cast<const __bf16*>(qkv.data_ptr()),
                        seql.data_ptr<int>(),
                        reinterpret_cast<__bf16*>(out.data_ptr()),
-                       lse.data_ptr<float>(), dmask.data_ptr<uint8_t>(), B, S,
-                       NH, p_q, scale,
+                       lse.data_ptr<float>(),
+                       reinterpret_cast<const uint32_t*>(
+                           dmask.data_ptr<int>()),
+                       B, S, NH, p_q, scale,
                        static_cast<uint64_t>(seed),
                        static_cast<uint64_t>(offset));
   } else {
@@ -845,10 +853,10 @@ torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
 
   dim3 grid((S + 127) / 128, B * NH), block(256);  // 128 keys per bwd block
   dim3 grid_dq((S + 127) / 128, B * NH);           // 128 q-rows per dq block
-  // K/V hold 128 rows; Q/Qt/dO/dOt 64 each; + 8 KB mask tile -> ~82 KB
+  // K/V hold 128 rows; Q/Qt/dO/dOt 64 each; + 1 KB bit-mask tile -> ~75 KB
   // (over the 64 KB default dynamic-LDS cap; MI355X has 160 KB per CU)
   const size_t lds = 8 * 64 * kStride * sizeof(__bf16) +
-                     2 * 64 * sizeof(float) + 64 * 128;
+                     2 * 64 * sizeof(float) + 64 * 4 * sizeof(uint32_t);
   const size_t lds_dq = 3 * 64 * kStride * sizeof(__bf16);
   HIP_CHECK(hipFuncSetAttribute(
       reinterpret_cast<const void*>(&attn_bwd_kernel<true>),
@@ -861,8 +869,9 @@ torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
   // same 8-bit quantization as the forward's mask generation, so the
   // backward keep-scale matches the stored mask's statistics exactly
   const float p_q = quantize_drop_p(p).second;
-  const uint8_t* mask_ptr =
-      train_drop ? dmask.data_ptr<uint8_t>() : nullptr;
+  const uint32_t* mask_ptr =
+      train_drop ? reinterpret_cast<const uint32_t*>(dmask.data_ptr<int>())
+                 : nullptr;
   auto args = [&](auto kernel, dim3 g, size_t lds_bytes) {
     hipLaunchKernelGGL(kernel, g, block, lds_bytes, stream,
                        reinterpret_cast<const __bf16*>(dout_c.data_ptr()),

@@ -630,7 +630,11 @@ def test_attention_dropout_numerics(B, S, NH):
     seqlens = torch.randint(S // 2, S + 1, (B,), device=DEV, dtype=torch.int32)
 
     out, lse, dmask = ext().attention_fwd(qkv, seqlens, NH, p, 77, 5)
-    keep = dmask.view(B, NH, S, S).float()
+    # dmask is bit-packed: [B*NH, S, ceil(S/32)] int32, LSB-first
+    bits = dmask.view(torch.uint8).reshape(B, NH, S, -1).int()
+    shifts = torch.arange(8, device=DEV, dtype=torch.int32)
+    keep = ((bits.unsqueeze(-1) >> shifts) & 1).reshape(B, NH, S, -1)
+    keep = keep[..., :S].float()
     # keep-rate sanity: mean within 3 sigma of 1 - p_q
     n = keep.numel()
     assert abs(keep.mean().item() - (1 - p_q)) < 4 * (p_q * (1 - p_q) / n) ** 0.5 + 1e-3
