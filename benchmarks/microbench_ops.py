@@ -136,6 +136,15 @@ def main():
     us = timeit(lambda: torch.autograd.grad(out, qkv, do, retain_graph=True))
     report("attn_bwd", us, 0,
            {"TFLOP_s": round(2.5 * flops / (us * 1e-6) / 1e12, 1)})
+    # production path: dropout p=0.1 (bit-packed stored mask: gen kernel
+    # in fwd, mask reads in fwd + both bwd kernels)
+    us = timeit(lambda: fused_attention(qkv.detach(), seqlens, NH, 0.1, True))
+    report("attn_fwd_drop", us, 0,
+           {"TFLOP_s": round(flops / (us * 1e-6) / 1e12, 1)})
+    out = fused_attention(qkv, seqlens, NH, 0.1, True)
+    us = timeit(lambda: torch.autograd.grad(out, qkv, do, retain_graph=True))
+    report("attn_bwd_drop", us, 0,
+           {"TFLOP_s": round(2.5 * flops / (us * 1e-6) / 1e12, 1)})
 
     # ---- cross entropy at vocab shape
     logits = torch.randn(rows, V, device=dev, dtype=dt)
