@@ -252,20 +252,23 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
     if (TRAIN_DROP) {
       // keep-mask pre-generated by dropout_mask_kernel (no RNG VALU
-      // work in the MFMA kernels); one 4-byte read per key-quad
+      // work in the MFMA kernels). k0 is a multiple of 64, so TWO
+      // 32-bit words cover this kv-tile's keys for the whole t-loop -
+      // quad extraction from registers replaces per-quad global loads.
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {
         const int q_abs = q_row[sub];
-        const bool q_ok = q_abs < S;
+        uint32_t mw[2] = {0xFFFFFFFFu, 0xFFFFFFFFu};
+        if (q_abs < S) {
+          const uint32_t* mrow = mask_base + static_cast<int64_t>(q_abs) * Sw;
+          const int w0 = k0 >> 5;
+          if (w0 < Sw) mw[0] = mrow[w0];
+          if (w0 + 1 < Sw) mw[1] = mrow[w0 + 1];
+        }
 #pragma unroll
         for (int t = 0; t < 4; ++t) {
-          const int key0 = k0 + t * 16 + g * 4;
-          uint32_t nib = 0xF;  // keys past the row end: keep (masked anyway)
-          if (q_ok && key0 < S)  // quads are word-aligned, never straddle
-            nib = (mask_base[static_cast<int64_t>(q_abs) * Sw +
-                             (key0 >> 5)] >>
-                   (key0 & 31)) &
-                  0xF;
+          const int kk = t * 16 + g * 4;  // key offset within the tile
+          const uint32_t nib = (mw[kk >> 5] >> (kk & 31)) & 0xF;
 #pragma unroll
           for (int j = 0; j < 4; ++j)
             sv[sub][t][j] = (nib >> j) & 1 ? sv[sub][t][j] * inv_keep : 0.f;
@@ -469,6 +472,21 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
       __bf16 e[8];
     };
     FB dsfrag[2][2];
+    // keep-mask words for this kv-tile (two words cover keys
+    // k0..k0+63 for each q-subtile; k0 is a multiple of 64)
+    uint32_t mw[2][2] = {{0xFFFFFFFFu, 0xFFFFFFFFu},
+                         {0xFFFFFFFFu, 0xFFFFFFFFu}};
+    if (TRAIN_DROP) {
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub)
+        if (q_row[sub] < S) {
+          const uint32_t* mrow =
+              mask_base + static_cast<int64_t>(q_row[sub]) * Sw;
+          const int w0 = k0 >> 5;
+          if (w0 < Sw) mw[sub][0] = mrow[w0];
+          if (w0 + 1 < Sw) mw[sub][1] = mrow[w0 + 1];
+        }
+    }
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
       const __bf16* krow = &K_lds[(t * 16 + li) * kStride];
@@ -485,15 +503,8 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
       }
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {
-        // read the forward's stored keep-mask: one word load per quad
-        uint32_t nib = 0xF;
-        if (TRAIN_DROP && q_row[sub] < S && k0 + t * 16 + g * 4 < S) {
-          const int key0 = k0 + t * 16 + g * 4;
-          nib = (mask_base[static_cast<int64_t>(q_row[sub]) * Sw +
-                           (key0 >> 5)] >>
-                 (key0 & 31)) &
-                0xF;
-        }
+        const int kk = t * 16 + g * 4;  // key offset within the tile
+        const uint32_t nib = (mw[sub][kk >> 5] >> (kk & 31)) & 0xF;
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int key = k0 + t * 16 + g * 4 + r;
