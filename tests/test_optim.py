@@ -9,6 +9,7 @@ import torch
 from bert_pytorch_amd.optim import (
     BertAdam,
     ConstantWarmUpScheduler,
+    CosineWarmUpScheduler,
     FusedAdam,
     FusedLAMB,
     LinearWarmUpScheduler,
@@ -149,6 +150,22 @@ def test_linear_constant_schedulers():
     for _ in range(5):
         c.step()
     assert opt2.param_groups[0]["lr"] == 1.0
+
+
+def test_cosine_scheduler_matches_reference_formula():
+    """Cosine decay reproduces the reference's exact (quirky) formula
+    lr * 0.5 * (1 + cos(pi + progress)) (src/schedulers.py:61-66) --
+    behavior parity matters more than the textbook curve."""
+    import math
+
+    opt = FusedAdam(_quadratic_problem(), lr=1.0)
+    sch = CosineWarmUpScheduler(opt, warmup=0.2, total_steps=10)
+    sch.step()  # last_epoch 0 -> 1: progress 0.1, in warmup
+    assert opt.param_groups[0]["lr"] == pytest.approx(0.1 / 0.2)
+    for _ in range(4):
+        sch.step()  # last_epoch -> 5: progress 0.5, past warmup
+    expected = 0.5 * (1.0 + math.cos(math.pi + 0.5))
+    assert opt.param_groups[0]["lr"] == pytest.approx(expected)
 
 
 def test_warmup_exp_decay_exp():
