@@ -89,6 +89,11 @@ def test_basic_tokenizer():
     assert bt.tokenize("Hello, World!") == ["hello", ",", "world", "!"]
     assert bt.tokenize("  a\tb\nc ") == ["a", "b", "c"]
     assert bt.tokenize("Café") == ["cafe"]  # accent stripped
+    # CJK chars split into single-char tokens (reference
+    # src/tokenization.py CJK padding semantics)
+    assert bt.tokenize("ab\u4e2d\u6587cd") == ["ab", "\u4e2d", "\u6587", "cd"]
+    # control chars removed, case preserved when lowercasing is off
+    assert BasicTokenizer(do_lower_case=False).tokenize("A\x00B") == ["AB"]
 
 
 def test_wordpiece_tokenizer_roundtrip(vocab_file):
