@@ -586,7 +586,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   __bf16* dOt_lds = dO_lds + 64 * kStride;           // [64][72] transposed
   float* lse_lds = reinterpret_cast<float*>(dOt_lds + 64 * kStride);  // [64]
   float* dlt_lds = lse_lds + 64;                                      // [64]
-  // keep-mask tile [64 q][128 keys] as bits, [64][4] words, staged per
+  // keep-mask tile [64 q][128 keys] as bits, [4 words][64 q], staged per
   // q-tile (raw global reads were byte columns - latency-bound at this
   // kernel's occupancy; 16 lanes share each word via LDS broadcast)
   uint32_t* mk_lds = reinterpret_cast<uint32_t*>(dlt_lds + 64);
@@ -674,7 +674,9 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
         uint32_t bits = 0xFFFFFFFFu;
         if (qrow_m < S && kw < Sw)
           bits = mask_base[static_cast<int64_t>(qrow_m) * Sw + kw];
-        mk_lds[mrow * 4 + mw] = bits;
+        // transposed [word][row] so a q-subtile's 4 row-words are one
+        // aligned b128 read in the recompute loop
+        mk_lds[mw * 64 + mrow] = bits;
       }
     }
     __syncthreads();
@@ -710,6 +712,10 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
         dp[sub] = MFMA16(da0, frag_row(vrow, 0, g), dp[sub]);
         dp[sub] = MFMA16(da1, frag_row(vrow, 32, g), dp[sub]);
       }
+      uint32_t mrow4[4];  // keep-mask words for rows mq*16+g*4+0..3
+      if (TRAIN_DROP)     // both key subtiles live in word column `wave`
+        *reinterpret_cast<uint4*>(mrow4) = *reinterpret_cast<const uint4*>(
+            &mk_lds[wave * 64 + mq * 16 + g * 4]);
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {
         const int key_abs = k0 + (sub ? key_local1 : key_local0);
@@ -726,10 +732,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
           float pkeep = pr;
           if (TRAIN_DROP) {
             const int key_l = sub ? key_local1 : key_local0;
-            const bool keep =
-                (mk_lds[(mq * 16 + g * 4 + r) * 4 + (key_l >> 5)] >>
-                 (key_l & 31)) &
-                1;
+            const bool keep = (mrow4[r] >> (key_l & 31)) & 1;
             dpd = keep ? dpd * inv_keep : 0.f;
             pkeep = keep ? pr * inv_keep : 0.f;  // dropped P feeds dV
           }
