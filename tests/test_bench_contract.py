@@ -65,3 +65,30 @@ def test_bench_json_contract(tmp_path):
     assert rec["config"]["seq_len"] == 128
     # global_batch = world * local_batch * accumulation
     assert rec["config"]["global_batch"] == 4
+
+
+@pytest.mark.timeout(300)
+def test_bench_two_rank_gloo(tmp_path):
+    """bench.py under the driver's exact torchrun launch pattern
+    (--nnodes=1 --nproc-per-node N --master-addr 127.0.0.1) with
+    world_size 2 on CPU/gloo: rank 0 prints one whole-job JSON line."""
+    cfg = tmp_path / "tiny.json"
+    cfg.write_text(json.dumps(TINY))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29518",
+         os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1",
+         "--local_batch", "2", "--accumulation", "2",
+         "--model_config", str(cfg)],
+        cwd=str(tmp_path), capture_output=True, text=True, timeout=280,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, "exactly one JSON line (rank 0 only)"
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["parallelism"] == "dp2"
+    # whole-job value: world * local_batch * 1e3 / ms_per_step
+    assert rec["value"] == pytest.approx(2 * 2 * 1e3 / rec["ms_per_step"], rel=0.05)
