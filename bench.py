@@ -175,7 +175,10 @@ def main():
         with torch.autocast(device.type, dtype=torch.bfloat16,
                             enabled=per_step_ac) if per_step_ac or not ac_enabled \
                 else contextlib.nullcontext():
-            scores, rel, glabels = model(ids, tt, mask, masked_lm_labels=labels)
+            scores, rel, glabels = model(
+                ids, tt, mask, masked_lm_labels=labels,
+                max_predictions_per_seq=phase["max_pred"],
+            )
             loss = criterion(scores, rel, glabels, nsp) / accum
         if sync or not isinstance(model, torch.nn.parallel.DistributedDataParallel):
             loss.backward()

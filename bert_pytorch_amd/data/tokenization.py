@@ -69,8 +69,13 @@ class BasicTokenizer:
     """Whitespace/punctuation/CJK splitting with optional lowercasing
     and accent stripping (reference semantics: src/tokenization.py:60-173)."""
 
-    def __init__(self, do_lower_case: bool = True):
+    NEVER_SPLIT = ("[UNK]", "[SEP]", "[PAD]", "[CLS]", "[MASK]")
+
+    def __init__(self, do_lower_case: bool = True, never_split=None):
         self.do_lower_case = do_lower_case
+        self.never_split = (
+            tuple(never_split) if never_split is not None else self.NEVER_SPLIT
+        )
 
     def tokenize(self, text: str) -> List[str]:
         text = self._clean(text)
@@ -78,6 +83,12 @@ class BasicTokenizer:
         tokens = text.strip().split()
         out: List[str] = []
         for token in tokens:
+            # special tokens pass through verbatim (reference:
+            # src/tokenization.py:64-65,74) so e.g. a literal "[UNK]" in
+            # raw SQuAD text is not lowercased/punct-split
+            if token in self.never_split:
+                out.append(token)
+                continue
             if self.do_lower_case:
                 token = token.lower()
                 token = self._strip_accents(token)

@@ -500,13 +500,44 @@ class BertForPreTraining(BertPreTrainedModel):
         token_type_ids=None,
         attention_mask=None,
         masked_lm_labels=None,
+        max_predictions_per_seq=None,
     ):
         encoded, pooled = self.bert(input_ids, token_type_ids, attention_mask)
         sequence_output = encoded[-1] if isinstance(encoded, list) else encoded
         if masked_lm_labels is None:
             return self.cls(sequence_output, pooled)
         flat_labels = masked_lm_labels.reshape(-1)
-        positions = torch.nonzero(flat_labels != -1, as_tuple=False).squeeze(-1)
+        if max_predictions_per_seq is not None:
+            # Sync-free padded gather: a fixed [B*max_pred] row budget so no
+            # torch.nonzero device->host count sync per micro-batch. Padding
+            # slots point at row 0 with label -1, which the CE loss ignores
+            # (zero gradient), so loss/grads match the variable-size gather
+            # and the reference's full [B,S,V] head exactly.
+            num = flat_labels.numel()
+            cap = masked_lm_labels.shape[0] * max_predictions_per_seq
+            mask = flat_labels != -1
+            slots = torch.cumsum(mask.to(torch.long), 0) - 1
+            slots = torch.where(
+                mask, slots.clamp_(max=cap), torch.full_like(slots, cap)
+            )
+            positions = torch.zeros(
+                cap + 1, dtype=torch.long, device=flat_labels.device
+            )
+            positions.scatter_(
+                0, slots, torch.arange(num, device=flat_labels.device)
+            )
+            gathered_labels = torch.full(
+                (cap + 1,), -1, dtype=flat_labels.dtype,
+                device=flat_labels.device,
+            )
+            gathered_labels.scatter_(0, slots, flat_labels)
+            positions = positions[:cap]
+            gathered_labels = gathered_labels[:cap]
+        else:
+            positions = torch.nonzero(
+                flat_labels != -1, as_tuple=False
+            ).squeeze(-1)
+            gathered_labels = flat_labels.index_select(0, positions)
         hidden = sequence_output.reshape(-1, sequence_output.shape[-1])
         masked_hidden = hidden.index_select(0, positions)
         scores = self.cls.predictions(masked_hidden)
@@ -515,7 +546,7 @@ class BertForPreTraining(BertPreTrainedModel):
             if self.cls.seq_relationship is not None and pooled is not None
             else None
         )
-        return scores, seq_rel, flat_labels.index_select(0, positions)
+        return scores, seq_rel, gathered_labels
 
 
 class BertForMaskedLM(BertPreTrainedModel):

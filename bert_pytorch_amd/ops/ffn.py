@@ -61,6 +61,13 @@ def ffn_supported(x: torch.Tensor) -> bool:
     global _aux_supported
     if not use_native(x):
         return False
+    # The epilogue GEMMs require bf16 inputs; outside autocast an fp32
+    # forward must take the standalone bias-GELU path instead (mirrors
+    # fused_attention's dtype gate).
+    if x.dtype != torch.bfloat16 and not (
+        x.is_cuda and torch.is_autocast_enabled()
+    ):
+        return False
     if _aux_supported is None:
         _aux_supported = bool(extension().gemm_gelu_aux_supported())
     return _aux_supported

@@ -92,6 +92,11 @@ class FusedAdam(Optimizer):
                     state["exp_avg_sq"] = torch.zeros_like(w, dtype=torch.float32)
                 state["step"] = step
                 if use_master:
+                    if "master" not in state:
+                        # resume from a checkpoint saved without masters
+                        # (e.g. fp32 run switched to --pure_bf16): lazily
+                        # seed the master from the param, like FusedLAMB
+                        state["master"] = p.detach().float().contiguous()
                     if "grad32" not in state:
                         state["grad32"] = torch.empty_like(state["master"])
                     lo_p.append(p)

@@ -40,6 +40,12 @@ class MetricLogger:
             os.makedirs(os.path.dirname(log_prefix) or ".", exist_ok=True)
             self._file = open(f"{log_prefix}.txt", "a", encoding="utf-8")
             self._csv_path = f"{log_prefix}_metrics.csv"
+            if os.path.exists(self._csv_path):
+                # Resume: adopt the existing header so appended rows line up.
+                with open(self._csv_path, newline="", encoding="utf-8") as f:
+                    header = f.readline().strip()
+                if header:
+                    self._csv_fields = header.split(",")
         self._tb = None
         if tensorboard_dir:
             try:
@@ -61,14 +67,28 @@ class MetricLogger:
         if self._csv_path:
             row: Dict[str, Any] = {"tag": tag, "step": step, **metrics}
             new_fields = [k for k in row if k not in self._csv_fields]
-            write_header = not os.path.exists(self._csv_path) or bool(new_fields)
-            self._csv_fields += new_fields
-            mode = "a" if os.path.exists(self._csv_path) and not new_fields else "a"
-            with open(self._csv_path, mode, newline="", encoding="utf-8") as f:
-                writer = csv.DictWriter(f, fieldnames=self._csv_fields)
-                if write_header:
+            exists = os.path.exists(self._csv_path)
+            if new_fields and exists and self._csv_fields:
+                # Schema widened mid-run (e.g. eval metrics joining a train
+                # log): rewrite the file under the union header so every row
+                # has the same columns and csv.DictReader parses it whole.
+                with open(self._csv_path, newline="", encoding="utf-8") as f:
+                    old_rows = list(csv.DictReader(f))
+                self._csv_fields += new_fields
+                with open(self._csv_path, "w", newline="", encoding="utf-8") as f:
+                    writer = csv.DictWriter(f, fieldnames=self._csv_fields)
                     writer.writeheader()
-                writer.writerow(row)
+                    writer.writerows(old_rows)
+                    writer.writerow(row)
+            else:
+                if new_fields:
+                    self._csv_fields += new_fields
+                write_header = not exists
+                with open(self._csv_path, "a", newline="", encoding="utf-8") as f:
+                    writer = csv.DictWriter(f, fieldnames=self._csv_fields)
+                    if write_header:
+                        writer.writeheader()
+                    writer.writerow(row)
         if self._tb:
             for key, value in metrics.items():
                 if isinstance(value, (int, float)):

@@ -182,6 +182,10 @@ torch::Tensor workspace_tensor(const torch::Device& dev) {
 void run_matmul(Plan& pl, const void* a, const void* b, void* d,
                 const void* bias, const void* aux, int64_t aux_ld,
                 void* workspace, hipStream_t stream) {
+  // The cached plan's descriptor is process-wide mutable state (bias/aux
+  // pointers); hold the lock across set+launch so two host threads
+  // sharing a shape cannot interleave pointer writes with the enqueue.
+  std::lock_guard<std::mutex> lock(g_mutex);
   set_pointers(pl, bias, aux, aux_ld);
   const float alpha = 1.f, beta = 0.f;
   BLT_CHECK(hipblasLtMatmul(handle(), pl.desc, &alpha, a, pl.la, b, pl.lb,
@@ -253,19 +257,16 @@ std::vector<torch::Tensor> gemm_dgelu_bgrad(torch::Tensor dout,
 // which cannot serve training because backward needs the
 // pre-activation); the fused-FFN path auto-disables via this check and
 // the standalone bias-GELU kernels run instead.
-bool gemm_gelu_aux_supported() {
-  static int cached = -1;
-  if (cached >= 0) return cached != 0;
+bool probe_epilogue(hipblasLtEpilogue_t epilogue, hipblasOperation_t opA) {
   hipblasLtMatmulDesc_t desc;
   if (hipblasLtMatmulDescCreate(&desc, HIPBLAS_COMPUTE_32F, HIP_R_32F) !=
       HIPBLAS_STATUS_SUCCESS) {
-    cached = 0;
     return false;
   }
-  int32_t ta = HIPBLAS_OP_T, tb = HIPBLAS_OP_N;
+  int32_t ta = opA, tb = HIPBLAS_OP_N;
   hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_TRANSA, &ta, 4);
   hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_TRANSB, &tb, 4);
-  uint32_t epi = HIPBLASLT_EPILOGUE_GELU_AUX_BIAS;
+  uint32_t epi = epilogue;
   hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi,
                                   4);
   void* dummy = reinterpret_cast<void*>(0x1000);
@@ -277,8 +278,12 @@ bool gemm_gelu_aux_supported() {
   int64_t ld = 4096;
   hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_LD,
                                   &ld, 8);
+  // Probe at the FFN shapes: m=4096 (=N), n=4096 (rows), k=1024 (=H).
+  // A col-major is [k,m] under OP_T and [m,k] under OP_N.
+  const int64_t a_rows = (opA == HIPBLAS_OP_T) ? 1024 : 4096;
+  const int64_t a_cols = (opA == HIPBLAS_OP_T) ? 4096 : 1024;
   hipblasLtMatrixLayout_t la, lb, lc;
-  hipblasLtMatrixLayoutCreate(&la, HIP_R_16BF, 1024, 4096, 1024);
+  hipblasLtMatrixLayoutCreate(&la, HIP_R_16BF, a_rows, a_cols, a_rows);
   hipblasLtMatrixLayoutCreate(&lb, HIP_R_16BF, 1024, 4096, 1024);
   hipblasLtMatrixLayoutCreate(&lc, HIP_R_16BF, 4096, 4096, 4096);
   hipblasLtMatmulPreference_t pref;
@@ -295,7 +300,20 @@ bool gemm_gelu_aux_supported() {
   hipblasLtMatrixLayoutDestroy(lb);
   hipblasLtMatrixLayoutDestroy(lc);
   hipblasLtMatmulDescDestroy(desc);
-  cached = found > 0 ? 1 : 0;
+  return found > 0;
+}
+
+bool gemm_gelu_aux_supported() {
+  static int cached = -1;
+  if (cached >= 0) return cached != 0;
+  // Training needs BOTH directions: a library that ships the forward
+  // AUX epilogue but not DGELU_BGRAD would otherwise enable the path
+  // and then fail mid-run in the first backward.
+  const bool fwd =
+      probe_epilogue(HIPBLASLT_EPILOGUE_GELU_AUX_BIAS, HIPBLAS_OP_T);
+  const bool bwd =
+      fwd && probe_epilogue(HIPBLASLT_EPILOGUE_DGELU_BGRAD, HIPBLAS_OP_N);
+  cached = (fwd && bwd) ? 1 : 0;
   return cached != 0;
 }
 

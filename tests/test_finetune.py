@@ -192,3 +192,50 @@ def test_run_ner_end_to_end(tmp_path, vocab_file, model_cfg, conll_file):
     ])
     run_ner.main(args)
     assert os.path.exists(tmp_path / "nerout" / "pytorch_model.bin")
+
+
+def test_basic_tokenizer_never_split():
+    """Special tokens pass through untouched (reference
+    src/tokenization.py:64-65,74)."""
+    bt = BasicTokenizer(do_lower_case=True)
+    assert bt.tokenize("foo [UNK] bar") == ["foo", "[UNK]", "bar"]
+    assert bt.tokenize("[CLS] Hi [SEP]") == ["[CLS]", "hi", "[SEP]"]
+    # default list covers all five BERT specials
+    for tok in ("[UNK]", "[SEP]", "[PAD]", "[CLS]", "[MASK]"):
+        assert bt.tokenize(f"x {tok} y") == ["x", tok, "y"]
+    # custom never_split overrides the default
+    bt2 = BasicTokenizer(do_lower_case=True, never_split=("<KEEP>",))
+    assert bt2.tokenize("a <KEEP> b") == ["a", "<KEEP>", "b"]
+    assert bt2.tokenize("a [UNK] b") == ["a", "[", "unk", "]", "b"]
+
+
+def test_metric_logger_csv_stable_schema(tmp_path):
+    """A train+eval log with different metric keys parses whole with
+    csv.DictReader (schema widening rewrites under a union header)."""
+    import csv as _csv
+
+    from bert_pytorch_amd.utils.logging import MetricLogger
+
+    prefix = str(tmp_path / "run")
+    ml = MetricLogger(log_prefix=prefix, verbose=False)
+    ml.log("train", 1, loss=1.5, lr=0.01)
+    ml.log("train", 2, loss=1.2, lr=0.01)
+    ml.log("eval", 2, f1=0.7)  # new field mid-file
+    ml.log("train", 3, loss=1.0, lr=0.009)
+    ml.close()
+    with open(prefix + "_metrics.csv", newline="") as f:
+        rows = list(_csv.DictReader(f))
+    assert len(rows) == 4
+    assert all(set(r.keys()) == {"tag", "step", "loss", "lr", "f1"} for r in rows)
+    assert rows[0]["loss"] == "1.5" and rows[2]["f1"] == "0.7"
+    assert rows[2]["loss"] == "" and rows[0]["f1"] == ""
+    # resume into the same file keeps one header and the schema
+    ml2 = MetricLogger(log_prefix=prefix, verbose=False)
+    ml2.log("train", 4, loss=0.9, lr=0.008)
+    ml2.close()
+    with open(prefix + "_metrics.csv", newline="") as f:
+        content = f.read()
+    assert content.count("tag,step") == 1
+    with open(prefix + "_metrics.csv", newline="") as f:
+        rows = list(_csv.DictReader(f))
+    assert len(rows) == 5 and rows[4]["loss"] == "0.9"

@@ -155,3 +155,40 @@ def test_activation_checkpointing_matches(tiny_config):
     with torch.no_grad():
         s2, _ = model_dropoutless(ids, tt, mask)
     torch.testing.assert_close(s1, s2)
+
+
+def test_padded_gather_matches_nonzero_gather(tiny_config):
+    """Sync-free padded max-pred gather = variable nonzero gather, in loss
+    AND gradients (padding rows carry label -1 -> zero CE gradient)."""
+    criterion = BertPretrainingCriterion(tiny_config.vocab_size)
+    ids, tt, mask, labels, nsp = _batch(tiny_config)
+
+    torch.manual_seed(7)
+    m1 = BertForPreTraining(tiny_config).train()
+    m2 = BertForPreTraining(tiny_config).train()
+    m2.load_state_dict(m1.state_dict())
+
+    torch.manual_seed(11)  # identical dropout draws for both forwards
+    s1, r1, l1 = m1(ids, tt, mask, masked_lm_labels=labels)
+    loss1 = criterion(s1, r1, l1, nsp)
+    loss1.backward()
+
+    torch.manual_seed(11)
+    s2, r2, l2 = m2(
+        ids, tt, mask, masked_lm_labels=labels, max_predictions_per_seq=5
+    )
+    # fixed row budget: bsz * max_pred
+    assert s2.shape[0] == ids.shape[0] * 5
+    loss2 = criterion(s2, r2, l2, nsp)
+    loss2.backward()
+
+    torch.testing.assert_close(loss1, loss2, rtol=1e-5, atol=1e-6)
+    for (n1, p1), (n2, p2) in zip(
+        m1.named_parameters(), m2.named_parameters()
+    ):
+        assert n1 == n2
+        if p1.grad is None:
+            assert p2.grad is None or torch.all(p2.grad == 0)
+            continue
+        torch.testing.assert_close(p1.grad, p2.grad, rtol=1e-4, atol=1e-6,
+                                   msg=lambda m, n=n1: f"{n}: {m}")
