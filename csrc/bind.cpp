@@ -45,6 +45,7 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor qkv,
                                          torch::Tensor seqlens,
                                          int64_t num_heads, double p,
                                          int64_t seed, int64_t offset);
+torch::Tensor tr16_probe(int64_t addr_mode);
 torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
                             torch::Tensor seqlens, torch::Tensor out,
                             torch::Tensor lse, torch::Tensor dmask,
@@ -100,6 +101,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &bpa::ce_bwd);
   m.def("attention_fwd", &bpa::attention_fwd);
   m.def("attention_bwd", &bpa::attention_bwd);
+  m.def("tr16_probe", &bpa::tr16_probe);
   m.def("multi_tensor_l2norm_sq", &bpa::multi_tensor_l2norm_sq);
   m.def("multi_tensor_clip_scale", &bpa::multi_tensor_clip_scale);
   m.def("fused_lamb", &bpa::fused_lamb);

@@ -53,6 +53,14 @@ typedef __hip_bfloat16 bf16_t;
 
 constexpr int kPad = 8;        // bf16 row pad: 72-element stride, bank-clean
 constexpr int kStride = 64 + kPad;
+// Stride for LDS images read with ds_read_b64_tr_b16: 80 elements (40
+// dwords) makes the hardware transpose read conflict-free within its
+// 32-lane groups ({40r+2c} for r 0..7, c 0..3 covers every even bank
+// exactly once). Row-fragment (b64) reads of the same image are 2-way
+// (rows r and r+8 share a bank) - acceptable where it saves a whole
+// second image (no stride satisfies both patterns: rows need
+// gcd(stride_dw,64)=4, tr needs gcd=8).
+constexpr int kTrStride = 80;
 
 // Build an A/B fragment from a row-major bf16 row pointer: elements at
 // k = base + g*4 + (e&3) + 16*(e>>2), loaded as two 8-byte chunks.
@@ -64,6 +72,35 @@ __device__ __forceinline__ bf16x8 frag_row(const __bf16* row, int base,
   } r;
   r.u[0] = *reinterpret_cast<const uint2*>(row + base + g * 4);
   r.u[1] = *reinterpret_cast<const uint2*>(row + base + 16 + g * 4);
+  return r.v;
+}
+
+typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4v;
+typedef __attribute__((address_space(3))) bf16x4v* lds_bf16x4p;
+
+// Transposed A/B fragment via gfx950's hardware transpose read, straight
+// from a NATURAL row-major [rows][kTrStride] bf16 LDS image: element
+// e = M[rb + g*4 + (e&3) + 16*(e>>2)][cb + li] - i.e. the fragment the
+// old code read from a separately-built transposed image. Semantics
+// probe-verified on hardware (tr16_probe): within each 16-lane group,
+// out[lane i][j] = mem[addr_of_lane(4j + (i>>2)) + (i&3)], so lane
+// (g, t) points at 4 contiguous elements of row rb + g*4 + (t>>2) and
+// the instruction delivers the group-transposed fragment. Replaces the
+// per-thread 16x ds_write_b16 transpose scatter at staging time.
+__device__ __forceinline__ bf16x8 frag_tr(const __bf16* img, int rb,
+                                          int cb) {
+  const int lane = threadIdx.x & 63;
+  const int g = (lane >> 4) & 3, t = lane & 15;
+  const __bf16* p0 =
+      img + (rb + g * 4 + (t >> 2)) * kTrStride + cb + (t & 3) * 4;
+  union {
+    bf16x8 v;
+    bf16x4v h[2];
+  } r;
+  r.h[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (lds_bf16x4p)(p0));
+  r.h[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (lds_bf16x4p)(p0 + 16 * kTrStride));
   return r.v;
 }
 
@@ -142,8 +179,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* K_lds = reinterpret_cast<__bf16*>(smem);             // [64][72]
-  __bf16* Vt_lds = K_lds + 64 * kStride;                       // [64][72]
-  float* alpha_lds = reinterpret_cast<float*>(Vt_lds + 64 * kStride);  // [4][2][16]
+  __bf16* V_lds = K_lds + 64 * kStride;   // [64][80] natural, tr-read
+  float* alpha_lds = reinterpret_cast<float*>(V_lds + 64 * kTrStride);  // [4][2][16]
   float* stat_lds = alpha_lds + 4 * 2 * 16;                    // [4][2][16]
 
   // this wave's two q-subtiles: rows q0 + wave*32 + sub*16 + li
@@ -175,20 +212,16 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
             reinterpret_cast<const uint4*>(kbase + static_cast<int64_t>(krow) * rs3 + colc);
         *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc]) = src[0];
         *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc + 8]) = src[1];
-        __bf16 vv[16];
         const uint4* vsrc =
             reinterpret_cast<const uint4*>(vbase + static_cast<int64_t>(krow) * rs3 + colc);
-        *reinterpret_cast<uint4*>(&vv[0]) = vsrc[0];
-        *reinterpret_cast<uint4*>(&vv[8]) = vsrc[1];
-#pragma unroll
-        for (int j = 0; j < 16; ++j) Vt_lds[(colc + j) * kStride + row] = vv[j];
+        *reinterpret_cast<uint4*>(&V_lds[row * kTrStride + colc]) = vsrc[0];
+        *reinterpret_cast<uint4*>(&V_lds[row * kTrStride + colc + 8]) = vsrc[1];
       } else {
         uint4 zero{0, 0, 0, 0};
         *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc]) = zero;
         *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc + 8]) = zero;
-#pragma unroll
-        for (int j = 0; j < 16; ++j)
-          Vt_lds[(colc + j) * kStride + row] = __bf16(0.f);
+        *reinterpret_cast<uint4*>(&V_lds[row * kTrStride + colc]) = zero;
+        *reinterpret_cast<uint4*>(&V_lds[row * kTrStride + colc + 8]) = zero;
       }
     }
     __syncthreads();
@@ -294,9 +327,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     }
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
-      const __bf16* vrow = &Vt_lds[(n * 16 + li) * kStride];
-      const bf16x8 vf0 = frag_row(vrow, 0, g);
-      const bf16x8 vf1 = frag_row(vrow, 32, g);
+      const bf16x8 vf0 = frag_tr(V_lds, 0, n * 16);
+      const bf16x8 vf1 = frag_tr(V_lds, 32, n * 16);
       acc_o[0][n] = MFMA16(pa[0][0], vf0, acc_o[0][n]);
       acc_o[1][n] = MFMA16(pa[1][0], vf0, acc_o[1][n]);
       acc_o[0][n] = MFMA16(pa[0][1], vf1, acc_o[0][n]);
@@ -404,9 +436,11 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
       TRAIN_DROP ? dmask + static_cast<int64_t>(bh) * S * Sw : nullptr;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  __bf16* K_lds = reinterpret_cast<__bf16*>(smem);   // [64][72] natural
-  __bf16* V_lds = K_lds + 64 * kStride;              // [64][72] natural
-  __bf16* Kt_lds = V_lds + 64 * kStride;             // [64][72] transposed
+  // K natural at kTrStride: serves BOTH the S^T-recompute row fragments
+  // (2-way bank conflicts, tolerated) and the dQ chain's K^T fragments
+  // via hardware transpose reads - no separate transposed image.
+  __bf16* K_lds = reinterpret_cast<__bf16*>(smem);   // [64][80] natural
+  __bf16* V_lds = K_lds + 64 * kTrStride;            // [64][72] natural
 
   // this wave's two q-subtiles: Q/dO fragments + lse/delta in registers
   int q_row[2];
@@ -438,29 +472,22 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
     {
       const int row = tid >> 2, colc = (tid & 3) * 16;
       const int krow = k0 + row;
-      __bf16 kv[16];
       if (krow < S) {
         const uint4* ks = reinterpret_cast<const uint4*>(
             kbase + static_cast<int64_t>(krow) * rs3 + colc);
-        *reinterpret_cast<uint4*>(&kv[0]) = ks[0];
-        *reinterpret_cast<uint4*>(&kv[8]) = ks[1];
+        *reinterpret_cast<uint4*>(&K_lds[row * kTrStride + colc]) = ks[0];
+        *reinterpret_cast<uint4*>(&K_lds[row * kTrStride + colc + 8]) = ks[1];
         const uint4* vs = reinterpret_cast<const uint4*>(
             vbase + static_cast<int64_t>(krow) * rs3 + colc);
         *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc]) = vs[0];
         *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc + 8]) = vs[1];
       } else {
         uint4 zero{0, 0, 0, 0};
-#pragma unroll
-        for (int j = 0; j < 16; ++j) kv[j] = __bf16(0.f);
+        *reinterpret_cast<uint4*>(&K_lds[row * kTrStride + colc]) = zero;
+        *reinterpret_cast<uint4*>(&K_lds[row * kTrStride + colc + 8]) = zero;
         *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc]) = zero;
         *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc + 8]) = zero;
       }
-      *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc]) =
-          *reinterpret_cast<uint4*>(&kv[0]);
-      *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc + 8]) =
-          *reinterpret_cast<uint4*>(&kv[8]);
-#pragma unroll
-      for (int j = 0; j < 16; ++j) Kt_lds[(colc + j) * kStride + row] = kv[j];
     }
     __syncthreads();
 
@@ -489,7 +516,7 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
     }
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
-      const __bf16* krow = &K_lds[(t * 16 + li) * kStride];
+      const __bf16* krow = &K_lds[(t * 16 + li) * kTrStride];
       const __bf16* vrow = &V_lds[(t * 16 + li) * kStride];
       const bf16x8 kf0 = frag_row(krow, 0, g), kf1 = frag_row(krow, 32, g);
       const bf16x8 vf0 = frag_row(vrow, 0, g), vf1 = frag_row(vrow, 32, g);
@@ -522,9 +549,8 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
     }
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
-      const __bf16* kt_row = &Kt_lds[(n * 16 + li) * kStride];
-      const bf16x8 ktf0 = frag_row(kt_row, 0, g);
-      const bf16x8 ktf1 = frag_row(kt_row, 32, g);
+      const bf16x8 ktf0 = frag_tr(K_lds, 0, n * 16);
+      const bf16x8 ktf1 = frag_tr(K_lds, 32, n * 16);
       acc_dq[0][n] = MFMA16(dsfrag[0][0].v, ktf0, acc_dq[0][n]);
       acc_dq[1][n] = MFMA16(dsfrag[1][0].v, ktf0, acc_dq[1][n]);
       acc_dq[0][n] = MFMA16(dsfrag[0][1].v, ktf1, acc_dq[0][n]);
@@ -578,13 +604,15 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       TRAIN_DROP ? dmask + static_cast<int64_t>(bh) * S * Sw : nullptr;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  // Q and dO at kTrStride serve BOTH row fragments (S/dP recompute;
+  // 2-way conflicts) and the dK/dV chains' transposed fragments via
+  // hardware transpose reads - the Qt/dOt images and their 16-scalar-
+  // write-per-thread transpose scatters are gone.
   __bf16* K_lds = reinterpret_cast<__bf16*>(smem);   // [128][72] natural
   __bf16* V_lds = K_lds + 128 * kStride;             // [128][72] natural
-  __bf16* Q_lds = V_lds + 128 * kStride;             // [64][72] natural
-  __bf16* Qt_lds = Q_lds + 64 * kStride;             // [64][72] transposed
-  __bf16* dO_lds = Qt_lds + 64 * kStride;            // [64][72] natural
-  __bf16* dOt_lds = dO_lds + 64 * kStride;           // [64][72] transposed
-  float* lse_lds = reinterpret_cast<float*>(dOt_lds + 64 * kStride);  // [64]
+  __bf16* Q_lds = V_lds + 128 * kStride;             // [64][80] natural
+  __bf16* dO_lds = Q_lds + 64 * kTrStride;           // [64][80] natural
+  float* lse_lds = reinterpret_cast<float*>(dO_lds + 64 * kTrStride);  // [64]
   float* dlt_lds = lse_lds + 64;                                      // [64]
   // keep-mask tile [64 q][128 keys] as bits, [4 words][64 q], staged per
   // q-tile (raw global reads were byte columns - latency-bound at this
@@ -632,33 +660,23 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     {
       const int row = tid >> 2, colc = (tid & 3) * 16;
       const int qrow = q0 + row;
-      __bf16 qv[16], dov[16];
+      uint4 qv0, qv1, dv0, dv1;
       if (qrow < S) {
         const uint4* qs =
             reinterpret_cast<const uint4*>(qbase + static_cast<int64_t>(qrow) * rs3 + colc);
-        *reinterpret_cast<uint4*>(&qv[0]) = qs[0];
-        *reinterpret_cast<uint4*>(&qv[8]) = qs[1];
+        qv0 = qs[0];
+        qv1 = qs[1];
         const uint4* ds =
             reinterpret_cast<const uint4*>(dobase + static_cast<int64_t>(qrow) * H + colc);
-        *reinterpret_cast<uint4*>(&dov[0]) = ds[0];
-        *reinterpret_cast<uint4*>(&dov[8]) = ds[1];
+        dv0 = ds[0];
+        dv1 = ds[1];
       } else {
-#pragma unroll
-        for (int j = 0; j < 16; ++j) qv[j] = dov[j] = __bf16(0.f);
+        qv0 = qv1 = dv0 = dv1 = uint4{0, 0, 0, 0};
       }
-      *reinterpret_cast<uint4*>(&Q_lds[row * kStride + colc]) =
-          *reinterpret_cast<uint4*>(&qv[0]);
-      *reinterpret_cast<uint4*>(&Q_lds[row * kStride + colc + 8]) =
-          *reinterpret_cast<uint4*>(&qv[8]);
-      *reinterpret_cast<uint4*>(&dO_lds[row * kStride + colc]) =
-          *reinterpret_cast<uint4*>(&dov[0]);
-      *reinterpret_cast<uint4*>(&dO_lds[row * kStride + colc + 8]) =
-          *reinterpret_cast<uint4*>(&dov[8]);
-#pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        Qt_lds[(colc + j) * kStride + row] = qv[j];
-        dOt_lds[(colc + j) * kStride + row] = dov[j];
-      }
+      *reinterpret_cast<uint4*>(&Q_lds[row * kTrStride + colc]) = qv0;
+      *reinterpret_cast<uint4*>(&Q_lds[row * kTrStride + colc + 8]) = qv1;
+      *reinterpret_cast<uint4*>(&dO_lds[row * kTrStride + colc]) = dv0;
+      *reinterpret_cast<uint4*>(&dO_lds[row * kTrStride + colc + 8]) = dv1;
       if (tid < 64) {
         const int qr = q0 + tid;
         lse_lds[tid] = (qr < S) ? lse[static_cast<int64_t>(bh) * S + qr] : 0.f;
@@ -697,8 +715,8 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     FB pfrag[2][2], dsfrag[2][2];
 #pragma unroll
     for (int mq = 0; mq < 4; ++mq) {
-      const __bf16* qrow_n = &Q_lds[(mq * 16 + li) * kStride];
-      const __bf16* dorow = &dO_lds[(mq * 16 + li) * kStride];
+      const __bf16* qrow_n = &Q_lds[(mq * 16 + li) * kTrStride];
+      const __bf16* dorow = &dO_lds[(mq * 16 + li) * kTrStride];
       const bf16x8 qa0 = frag_row(qrow_n, 0, g), qa1 = frag_row(qrow_n, 32, g);
       const bf16x8 da0 = frag_row(dorow, 0, g), da1 = frag_row(dorow, 32, g);
       f32x4 acc[2] = {}, dp[2] = {};
@@ -746,10 +764,10 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     // dV^T += dO^T x P ; dK^T += Q^T x dS — dO^T/Q^T fragments shared
 #pragma unroll
     for (int m = 0; m < 4; ++m) {
-      const __bf16* dot_row = &dOt_lds[(m * 16 + li) * kStride];
-      const __bf16* qt_row = &Qt_lds[(m * 16 + li) * kStride];
-      const bf16x8 dt0 = frag_row(dot_row, 0, g), dt1 = frag_row(dot_row, 32, g);
-      const bf16x8 qt0 = frag_row(qt_row, 0, g), qt1 = frag_row(qt_row, 32, g);
+      const bf16x8 dt0 = frag_tr(dO_lds, 0, m * 16);
+      const bf16x8 dt1 = frag_tr(dO_lds, 32, m * 16);
+      const bf16x8 qt0 = frag_tr(Q_lds, 0, m * 16);
+      const bf16x8 qt1 = frag_tr(Q_lds, 32, m * 16);
       dv_acc[0][m] = MFMA16(dt0, pfrag[0][0].v, dv_acc[0][m]);
       dv_acc[1][m] = MFMA16(dt0, pfrag[1][0].v, dv_acc[1][m]);
       dv_acc[0][m] = MFMA16(dt1, pfrag[0][1].v, dv_acc[0][m]);
@@ -781,6 +799,42 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// ds_read_b64_tr_b16 semantics probe (debug utility; used once per ROCm
+// bump to verify the lane->element mapping the PV/dK/dV fragment reads
+// rely on). LDS is filled with the identity pattern (element i = i) and
+// each lane reads at its own 8-byte-aligned address (8*lane bytes);
+// out[l][j] then IS the element index lane l's slot j received.
+// ---------------------------------------------------------------------------
+__global__ void tr16_probe_kernel(float* __restrict__ out, int addr_mode) {
+  __shared__ __bf16 lds[1024];
+  const int t = threadIdx.x;
+  for (int i = t; i < 1024; i += 64) lds[i] = __bf16(static_cast<float>(i));
+  __syncthreads();
+  typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4v;
+  int elem_off;
+  switch (addr_mode) {
+    case 0: elem_off = t * 4; break;           // 8 B per lane, contiguous
+    case 1: elem_off = 0; break;               // uniform base
+    case 2: elem_off = (t & 15) * 4; break;    // repeats per 16-lane group
+    default: elem_off = (t >> 4) * 4; break;   // group-constant
+  }
+  auto p = (__attribute__((address_space(3))) bf16x4v*)(&lds[elem_off]);
+  bf16x4v v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) out[t * 4 + j] = static_cast<float>(v[j]);
+}
+
+torch::Tensor tr16_probe(int64_t addr_mode) {
+  auto out = torch::empty({64, 4}, torch::TensorOptions()
+                                       .dtype(torch::kFloat32)
+                                       .device(torch::kCUDA));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     out.data_ptr<float>(), static_cast<int>(addr_mode));
+  return out;
+}
+
+// ---------------------------------------------------------------------------
 // host wrappers
 // ---------------------------------------------------------------------------
 std::vector<torch::Tensor> attention_fwd(torch::Tensor qkv,
@@ -809,8 +863,8 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor qkv,
   const float scale = 1.0f / sqrtf(64.f);
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid((S + 127) / 128, B * NH), block(256);  // 128 q-rows per block
-  const size_t lds =
-      2 * 64 * kStride * sizeof(__bf16) + 4 * 64 * sizeof(float);
+  const size_t lds = (64 * kStride + 64 * kTrStride) * sizeof(__bf16) +
+                     4 * 64 * sizeof(float);
   if (train_drop) {
     const auto [thresh, p_q] = quantize_drop_p(p);
     const int64_t words = static_cast<int64_t>(B) * NH * S * Sw;
@@ -867,11 +921,14 @@ torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
 
   dim3 grid((S + 127) / 128, B * NH), block(256);  // 128 keys per bwd block
   dim3 grid_dq((S + 127) / 128, B * NH);           // 128 q-rows per dq block
-  // K/V hold 128 rows; Q/Qt/dO/dOt 64 each; + 1 KB bit-mask tile -> ~75 KB
-  // (over the 64 KB default dynamic-LDS cap; MI355X has 160 KB per CU)
-  const size_t lds = 8 * 64 * kStride * sizeof(__bf16) +
+  // K/V hold 128 rows at kStride; Q/dO 64 rows at kTrStride (tr-read);
+  // + 1 KB bit-mask tile -> ~59 KB (over the 64 KB default dynamic-LDS
+  // cap; MI355X has 160 KB per CU)
+  const size_t lds = (4 * 64 * kStride + 2 * 64 * kTrStride) *
+                         sizeof(__bf16) +
                      2 * 64 * sizeof(float) + 64 * 4 * sizeof(uint32_t);
-  const size_t lds_dq = 3 * 64 * kStride * sizeof(__bf16);
+  const size_t lds_dq =
+      (64 * kStride + 64 * kTrStride) * sizeof(__bf16);
   HIP_CHECK(hipFuncSetAttribute(
       reinterpret_cast<const void*>(&attn_bwd_kernel<true>),
       hipFuncAttributeMaxDynamicSharedMemorySize, lds));
