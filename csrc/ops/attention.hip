@@ -228,6 +228,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
     // S^T tiles: one K-fragment pair feeds both subtiles' MFMA chains
     f32x4 s_acc[2][4];
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
       const __bf16* krow = &K_lds[(t * 16 + li) * kStride];
@@ -241,6 +242,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       s_acc[0][t] = a0;
       s_acc[1][t] = a1;
     }
+    __builtin_amdgcn_s_setprio(0);
 
     float sv[2][4][4];
 #pragma unroll
@@ -325,6 +327,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         pa[sub][c] = pk.v;
       }
     }
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
       const bf16x8 vf0 = frag_tr(V_lds, 0, n * 16);
@@ -334,6 +337,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       acc_o[0][n] = MFMA16(pa[0][1], vf1, acc_o[0][n]);
       acc_o[1][n] = MFMA16(pa[1][1], vf1, acc_o[1][n]);
     }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   // epilogue: normalize rows by l (broadcast per-wave through LDS)
@@ -499,6 +503,7 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
       __bf16 e[8];
     };
     FB dsfrag[2][2];
+    // (setprio brackets are inside the loops below)
     // keep-mask words for this kv-tile (two words cover keys
     // k0..k0+63 for each q-subtile; k0 is a multiple of 64)
     uint32_t mw[2][2] = {{0xFFFFFFFFu, 0xFFFFFFFFu},
@@ -547,6 +552,7 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
         }
       }
     }
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
       const bf16x8 ktf0 = frag_tr(K_lds, 0, n * 16);
@@ -556,6 +562,7 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
       acc_dq[0][n] = MFMA16(dsfrag[0][1].v, ktf1, acc_dq[0][n]);
       acc_dq[1][n] = MFMA16(dsfrag[1][1].v, ktf1, acc_dq[1][n]);
     }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   // epilogue: one store per element into dqkv Q slots
@@ -652,52 +659,63 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   f32x4 dv_acc[2][4] = {};  // dV^T[sub][dh-tile] for the wave's 2x16 keys
   f32x4 dk_acc[2][4] = {};
 
+  // async-STAGE split (T14): each q-tile's Q/dO/lse/delta/mask global
+  // loads are ISSUED one iteration early, so their HBM latency hides
+  // under the previous tile's 64-MFMA recompute instead of sitting
+  // serialized between the two staging barriers. The +16 VGPRs of
+  // in-flight data keep this kernel at its LDS-bound 2 waves/SIMD.
+  const int st_row = tid >> 2, st_colc = (tid & 3) * 16;
+  uint4 pf_q0, pf_q1, pf_d0, pf_d1;
+  auto issue_loads = [&](int q0t) {
+    const int qrow = q0t + st_row;
+    if (qrow < S) {
+      const uint4* qs = reinterpret_cast<const uint4*>(
+          qbase + static_cast<int64_t>(qrow) * rs3 + st_colc);
+      pf_q0 = qs[0];
+      pf_q1 = qs[1];
+      const uint4* ds = reinterpret_cast<const uint4*>(
+          dobase + static_cast<int64_t>(qrow) * H + st_colc);
+      pf_d0 = ds[0];
+      pf_d1 = ds[1];
+    } else {
+      pf_q0 = pf_q1 = pf_d0 = pf_d1 = uint4{0, 0, 0, 0};
+    }
+  };
+  issue_loads(0);
+
   const int n_q = (S + 63) / 64;
   for (int qt = 0; qt < n_q; ++qt) {
     const int q0 = qt * 64;
-    __syncthreads();
-    // stage Q, Q^T, dO, dO^T + lse/delta for this q-tile
+    __syncthreads();  // all waves done reading the previous tile's Q/dO
+    // stage this q-tile from the prefetched registers
     {
-      const int row = tid >> 2, colc = (tid & 3) * 16;
-      const int qrow = q0 + row;
-      uint4 qv0, qv1, dv0, dv1;
-      if (qrow < S) {
-        const uint4* qs =
-            reinterpret_cast<const uint4*>(qbase + static_cast<int64_t>(qrow) * rs3 + colc);
-        qv0 = qs[0];
-        qv1 = qs[1];
-        const uint4* ds =
-            reinterpret_cast<const uint4*>(dobase + static_cast<int64_t>(qrow) * H + colc);
-        dv0 = ds[0];
-        dv1 = ds[1];
-      } else {
-        qv0 = qv1 = dv0 = dv1 = uint4{0, 0, 0, 0};
-      }
-      *reinterpret_cast<uint4*>(&Q_lds[row * kTrStride + colc]) = qv0;
-      *reinterpret_cast<uint4*>(&Q_lds[row * kTrStride + colc + 8]) = qv1;
-      *reinterpret_cast<uint4*>(&dO_lds[row * kTrStride + colc]) = dv0;
-      *reinterpret_cast<uint4*>(&dO_lds[row * kTrStride + colc + 8]) = dv1;
+      *reinterpret_cast<uint4*>(&Q_lds[st_row * kTrStride + st_colc]) = pf_q0;
+      *reinterpret_cast<uint4*>(&Q_lds[st_row * kTrStride + st_colc + 8]) =
+          pf_q1;
+      *reinterpret_cast<uint4*>(&dO_lds[st_row * kTrStride + st_colc]) =
+          pf_d0;
+      *reinterpret_cast<uint4*>(&dO_lds[st_row * kTrStride + st_colc + 8]) =
+          pf_d1;
       if (tid < 64) {
         const int qr = q0 + tid;
         lse_lds[tid] = (qr < S) ? lse[static_cast<int64_t>(bh) * S + qr] : 0.f;
         dlt_lds[tid] = (qr < S) ? delta[static_cast<int64_t>(bh) * S + qr] : 0.f;
       }
       if (TRAIN_DROP) {
-        // stage the keep-mask tile [64 q][128 keys] = [64][4] words:
-        // one word per thread; words past the row end fill with ones
-        // (those keys are already zeroed by the kvalid/slen guards)
-        const int mrow = tid >> 2, mw = tid & 3;
-        const int qrow_m = q0 + mrow;
-        const int kw = (k0 >> 5) + mw;  // k0 is a multiple of 128
+        // keep-mask tile [64 q][128 keys] bits, transposed [word][row]
+        // so a q-subtile's 4 row-words are one aligned b128 read below;
+        // words past the row end are ones (those keys are zeroed by the
+        // kvalid/slen guards)
+        const int kw = (k0 >> 5) + (tid & 3);  // k0 is a multiple of 128
+        const int qrow_m = q0 + st_row;
         uint32_t bits = 0xFFFFFFFFu;
         if (qrow_m < S && kw < Sw)
           bits = mask_base[static_cast<int64_t>(qrow_m) * Sw + kw];
-        // transposed [word][row] so a q-subtile's 4 row-words are one
-        // aligned b128 read in the recompute loop
-        mk_lds[mw * 64 + mrow] = bits;
+        mk_lds[(tid & 3) * 64 + st_row] = bits;
       }
     }
     __syncthreads();
+    if (qt + 1 < n_q) issue_loads(q0 + 64);
 
     // recompute S tiles for the wave's 2x16 keys x 64 q-rows:
     // S[q][key]: A = Q rows, B = K^T (natural K rows); C col = key = li.
@@ -762,6 +780,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     }
 
     // dV^T += dO^T x P ; dK^T += Q^T x dS — dO^T/Q^T fragments shared
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int m = 0; m < 4; ++m) {
       const bf16x8 dt0 = frag_tr(dO_lds, 0, m * 16);
@@ -777,6 +796,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       dk_acc[0][m] = MFMA16(qt1, dsfrag[0][1].v, dk_acc[0][m]);
       dk_acc[1][m] = MFMA16(qt1, dsfrag[1][1].v, dk_acc[1][m]);
     }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   // write dK/dV straight into dqkv (this block owns keys k0..k0+127)
