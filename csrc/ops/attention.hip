@@ -182,6 +182,10 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   __bf16* V_lds = K_lds + 64 * kStride;   // [64][80] natural, tr-read
   float* alpha_lds = reinterpret_cast<float*>(V_lds + 64 * kTrStride);  // [4][2][16]
   float* stat_lds = alpha_lds + 4 * 2 * 16;                    // [4][2][16]
+  // keep-mask words for the current kv-tile, [128 q][2 words]: staged
+  // coalesced (one word per thread) instead of per-lane scattered 4-byte
+  // global reads in the softmax loop
+  uint32_t* mk_lds = reinterpret_cast<uint32_t*>(stat_lds + 4 * 2 * 16);
 
   // this wave's two q-subtiles: rows q0 + wave*32 + sub*16 + li
   int q_row[2];
@@ -222,6 +226,14 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         *reinterpret_cast<uint4*>(&K_lds[row * kStride + colc + 8]) = zero;
         *reinterpret_cast<uint4*>(&V_lds[row * kTrStride + colc]) = zero;
         *reinterpret_cast<uint4*>(&V_lds[row * kTrStride + colc + 8]) = zero;
+      }
+      if (TRAIN_DROP) {
+        const int qrow_m = q0 + (tid >> 1);
+        const int kw = (k0 >> 5) + (tid & 1);  // 2 words cover 64 keys
+        uint32_t bits = 0xFFFFFFFFu;
+        if (qrow_m < S && kw < Sw)
+          bits = mask_base[static_cast<int64_t>(qrow_m) * Sw + kw];
+        mk_lds[tid] = bits;
       }
     }
     __syncthreads();
@@ -289,17 +301,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       // keep-mask pre-generated by dropout_mask_kernel (no RNG VALU
       // work in the MFMA kernels). k0 is a multiple of 64, so TWO
       // 32-bit words cover this kv-tile's keys for the whole t-loop -
-      // quad extraction from registers replaces per-quad global loads.
+      // one b64 LDS read per subtile replaces per-lane global loads.
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {
-        const int q_abs = q_row[sub];
-        uint32_t mw[2] = {0xFFFFFFFFu, 0xFFFFFFFFu};
-        if (q_abs < S) {
-          const uint32_t* mrow = mask_base + static_cast<int64_t>(q_abs) * Sw;
-          const int w0 = k0 >> 5;
-          if (w0 < Sw) mw[0] = mrow[w0];
-          if (w0 + 1 < Sw) mw[1] = mrow[w0 + 1];
-        }
+        const int q_loc = wave * 32 + sub * 16 + li;
+        uint32_t mw[2];
+        *reinterpret_cast<uint2*>(mw) =
+            *reinterpret_cast<const uint2*>(&mk_lds[q_loc * 2]);
 #pragma unroll
         for (int t = 0; t < 4; ++t) {
           const int kk = t * 16 + g * 4;  // key offset within the tile
@@ -445,6 +453,7 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
   // via hardware transpose reads - no separate transposed image.
   __bf16* K_lds = reinterpret_cast<__bf16*>(smem);   // [64][80] natural
   __bf16* V_lds = K_lds + 64 * kTrStride;            // [64][72] natural
+  uint32_t* mk_lds = reinterpret_cast<uint32_t*>(V_lds + 64 * kStride);
 
   // this wave's two q-subtiles: Q/dO fragments + lse/delta in registers
   int q_row[2];
@@ -492,6 +501,14 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
         *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc]) = zero;
         *reinterpret_cast<uint4*>(&V_lds[row * kStride + colc + 8]) = zero;
       }
+      if (TRAIN_DROP) {
+        const int qrow_m = q0 + (tid >> 1);
+        const int kw = (k0 >> 5) + (tid & 1);
+        uint32_t bits = 0xFFFFFFFFu;
+        if (qrow_m < S && kw < Sw)
+          bits = mask_base[static_cast<int64_t>(qrow_m) * Sw + kw];
+        mk_lds[tid] = bits;
+      }
     }
     __syncthreads();
 
@@ -510,14 +527,11 @@ __global__ __launch_bounds__(256) void attn_dq_kernel(
                          {0xFFFFFFFFu, 0xFFFFFFFFu}};
     if (TRAIN_DROP) {
 #pragma unroll
-      for (int sub = 0; sub < 2; ++sub)
-        if (q_row[sub] < S) {
-          const uint32_t* mrow =
-              mask_base + static_cast<int64_t>(q_row[sub]) * Sw;
-          const int w0 = k0 >> 5;
-          if (w0 < Sw) mw[sub][0] = mrow[w0];
-          if (w0 + 1 < Sw) mw[sub][1] = mrow[w0 + 1];
-        }
+      for (int sub = 0; sub < 2; ++sub) {
+        const int q_loc = wave * 32 + sub * 16 + li;
+        *reinterpret_cast<uint2*>(mw[sub]) =
+            *reinterpret_cast<const uint2*>(&mk_lds[q_loc * 2]);
+      }
     }
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
@@ -884,7 +898,7 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor qkv,
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid((S + 127) / 128, B * NH), block(256);  // 128 q-rows per block
   const size_t lds = (64 * kStride + 64 * kTrStride) * sizeof(__bf16) +
-                     4 * 64 * sizeof(float);
+                     4 * 64 * sizeof(float) + 128 * 2 * sizeof(uint32_t);
   if (train_drop) {
     const auto [thresh, p_q] = quantize_drop_p(p);
     const int64_t words = static_cast<int64_t>(B) * NH * S * Sw;
@@ -947,8 +961,8 @@ torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
   const size_t lds = (4 * 64 * kStride + 2 * 64 * kTrStride) *
                          sizeof(__bf16) +
                      2 * 64 * sizeof(float) + 64 * 4 * sizeof(uint32_t);
-  const size_t lds_dq =
-      (64 * kStride + 64 * kTrStride) * sizeof(__bf16);
+  const size_t lds_dq = (64 * kStride + 64 * kTrStride) * sizeof(__bf16) +
+                        128 * 2 * sizeof(uint32_t);
   HIP_CHECK(hipFuncSetAttribute(
       reinterpret_cast<const void*>(&attn_bwd_kernel<true>),
       hipFuncAttributeMaxDynamicSharedMemorySize, lds));
