@@ -69,6 +69,17 @@ def parse_args():
                    help="bf16 model weights + fp32 LAMB masters instead of "
                         "fp32 weights + autocast (no per-microbatch weight "
                         "casts; bf16 gradient all-reduce)")
+    p.add_argument("--fp32_weights", action="store_true",
+                   help="fp32 weights + per-microbatch autocast casts (the "
+                        "pre-round-2 default)")
+    p.add_argument("--bf16_weights", action="store_true", default=True,
+                   help="bf16 MATMUL/embedding weights only (LN params and "
+                        "biases stay fp32) under bf16 autocast: the per-"
+                        "microbatch big-weight casts become no-ops and their "
+                        "grads accumulate in bf16 at half the bytes, while "
+                        "the fused kernels keep their fp32-param call "
+                        "pattern (avoids pure_bf16's per-call grad "
+                        "downcasts). fp32 LAMB masters for the bf16 params.")
     p.add_argument("--seed", type=int, default=1234)
     return p.parse_args()
 
@@ -118,8 +129,16 @@ def main():
     if config.vocab_size % 64:
         config.vocab_size += 64 - config.vocab_size % 64
     model = BertForPreTraining(config).to(device)
+    if args.fp32_weights or args.pure_bf16:
+        args.bf16_weights = False
     if args.pure_bf16:
         model = model.to(torch.bfloat16)
+    elif args.bf16_weights and use_cuda:
+        # CUDA-only mode (CPU smoke runs keep fp32: no autocast to
+        # coordinate mixed dtypes there)
+        for n, prm in model.named_parameters():
+            if prm.dim() >= 2 and "LayerNorm" not in n:
+                prm.data = prm.data.to(torch.bfloat16)
     criterion = BertPretrainingCriterion(config.vocab_size)
     model = comm.wrap_ddp(model, local_rank,
                           grad_compress=args.grad_compress)
@@ -133,7 +152,7 @@ def main():
              "weight_decay": 0.0},
         ],
         lr=phase["lr"],
-        master_weights=args.pure_bf16,
+        master_weights=args.pure_bf16 or args.bf16_weights,
     )
     scheduler = PolyWarmUpScheduler(optimizer, warmup=0.2843, total_steps=7038)
     gen = torch.Generator(device=device).manual_seed(args.seed + rank)
@@ -240,9 +259,12 @@ def main():
                 "local_batch": bsz,
                 "accumulation": accum,
                 "optimizer": "FusedLAMB (HIP multi-tensor)"
-                + (" + fp32 masters" if args.pure_bf16 else ""),
+                + (" + fp32 masters"
+                   if args.pure_bf16 or args.bf16_weights else ""),
                 "weights": "bf16 (fp32 LAMB masters)" if args.pure_bf16
-                else "fp32 + bf16 autocast",
+                else ("bf16 matmul weights + fp32 LN/bias (fp32 LAMB "
+                      "masters)" if args.bf16_weights
+                      else "fp32 + bf16 autocast"),
                 "parallelism": f"dp{world}",
             },
         }

@@ -36,6 +36,7 @@ def main():
     ap.add_argument("--phase", type=int, default=1, choices=[1, 2])
     ap.add_argument("--steps", type=int, default=3, help="accum windows")
     ap.add_argument("--accumulation", type=int, default=8)
+    ap.add_argument("--pure_bf16", action="store_true")
     args = ap.parse_args()
     from bert_pytorch_amd.utils import tunable
 
@@ -48,6 +49,8 @@ def main():
     if config.vocab_size % 64:
         config.vocab_size += 64 - config.vocab_size % 64
     model = BertForPreTraining(config).to(device)
+    if args.pure_bf16:
+        model = model.to(torch.bfloat16)
     criterion = BertPretrainingCriterion(config.vocab_size)
     named = list(model.named_parameters())
     no_decay = ("bias", "LayerNorm", "qkv_bias")
@@ -60,6 +63,7 @@ def main():
              "weight_decay": 0.0},
         ],
         lr=1e-3,
+        master_weights=args.pure_bf16,
     )
     bsz, seq, mp = phase["local_batch"], phase["seq_len"], phase["max_pred"]
     gen = torch.Generator(device=device).manual_seed(0)
@@ -72,8 +76,11 @@ def main():
     nsp = torch.randint(0, 2, (bsz,), generator=gen, device=device)
     model.train()
 
+    import contextlib
+
     def micro(i):
-        with torch.autocast("cuda", dtype=torch.bfloat16):
+        with torch.autocast("cuda", dtype=torch.bfloat16) \
+                if not args.pure_bf16 else contextlib.nullcontext():
             s, r, gl = model(ids, tt, mask, masked_lm_labels=labels,
                              max_predictions_per_seq=mp)
             loss = criterion(s, r, gl, nsp) / args.accumulation

@@ -67,6 +67,13 @@ def parse_arguments(args=None) -> argparse.Namespace:
                         choices=["lamb", "adam"],
                         help="fused LAMB (BERT two-phase) or fused Adam "
                              "(RoBERTa single-phase)")
+    parser.add_argument("--bf16_weights", action="store_true",
+                        help="bf16 matmul/embedding weights + fp32 LN/bias "
+                             "under bf16 autocast, fp32 optimizer masters "
+                             "for the bf16 params: the per-microbatch "
+                             "big-weight autocast casts become no-ops and "
+                             "their grads accumulate in bf16 (measured "
+                             "+2-3%% seq/s over fp32 weights on MI355X)")
     parser.add_argument("--pure_bf16", action="store_true",
                         help="bf16 model weights + fp32 optimizer masters "
                              "instead of fp32 weights + autocast")
@@ -101,7 +108,7 @@ def parse_arguments(args=None) -> argparse.Namespace:
                         default=int(os.environ.get("LOCAL_RANK", 0)))
     ns = merge_config_and_args(parser, args)
     # bool-ish JSON values
-    for key in ("fp16", "bf16", "kfac", "disable_progress_bar",
+    for key in ("fp16", "bf16", "bf16_weights", "kfac", "disable_progress_bar",
                 "checkpoint_activations"):
         setattr(ns, key, bool(getattr(ns, key)))
     return ns
@@ -175,6 +182,10 @@ def prepare_model(args, device):
     model.to(device)
     if args.pure_bf16:
         model.to(torch.bfloat16)
+    elif args.bf16_weights and device.type == "cuda":
+        for n, prm in model.named_parameters():
+            if prm.dim() >= 2 and "LayerNorm" not in n:
+                prm.data = prm.data.to(torch.bfloat16)
     if args.checkpoint_activations:
         model.checkpoint_activations(True)
     global_steps = max(0, resume_step - args.previous_phase_end_step)
@@ -196,11 +207,13 @@ def prepare_optimizers(args, model, resume_state):
         from bert_pytorch_amd.optim import FusedAdam  # noqa: PLC0415
 
         optimizer = FusedAdam(
-            groups, lr=args.learning_rate, master_weights=args.pure_bf16
+            groups, lr=args.learning_rate,
+            master_weights=args.pure_bf16 or args.bf16_weights,
         )
     else:
         optimizer = FusedLAMB(
-            groups, lr=args.learning_rate, master_weights=args.pure_bf16
+            groups, lr=args.learning_rate,
+            master_weights=args.pure_bf16 or args.bf16_weights,
         )
     from bert_pytorch_amd.optim import (  # noqa: PLC0415
         ConstantWarmUpScheduler,
