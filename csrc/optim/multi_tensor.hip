@@ -9,12 +9,17 @@
 //   * fused_adam             : one-stage AdamW/Adam apply
 //
 // A chunk table ([tensor_idx, element_offset] per 64Ki-element chunk)
-// and per-tensor pointer tables are built host-side each call and
-// copied once; all kernels then run without host synchronization (the
-// LAMB trust ratio and the clip factor are read on-device).
+// and per-tensor pointer/size tables are built host-side ONCE per
+// distinct tensor-list (cached by pointer+size key; steady-state steps
+// do zero H2D metadata copies); all kernels run without host
+// synchronization (the LAMB trust ratio and the clip factor are read
+// on-device).
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+
+#include <mutex>
+#include <unordered_map>
 
 #include "../common.h"
 
@@ -28,20 +33,54 @@ namespace {
 struct ChunkedList {
   torch::Tensor ptrs;    // [n_lists, n_tensors] int64 (device)
   torch::Tensor chunks;  // [n_chunks, 3] int32: tensor, offset_lo, offset_hi
+  torch::Tensor sizes;   // [n_tensors] int64 (device)
   int n_chunks;
 };
+
+// Metadata cache: the chunk/pointer/size device tables depend only on
+// the tensors' data pointers and sizes, which are stable across steps
+// (the caching allocator reuses blocks), so building the ~400-tensor
+// tables host-side and copying them H2D every optimizer step is wasted
+// per-step latency. Key = FNV-1a over (n_lists,
+// every data_ptr, every numel); kernels consume only pointers+sizes, so
+// a key hit is correct by construction even if the tensor OBJECTS
+// changed (e.g. set_to_none grads re-allocated at the same address).
+std::mutex g_meta_mutex;
+std::unordered_map<uint64_t, ChunkedList> g_meta_cache;
 
 ChunkedList build_chunks(const std::vector<std::vector<torch::Tensor>>& lists,
                          const torch::Device& device) {
   const int n_lists = lists.size();
   const int n = lists[0].size();
+
+  uint64_t key = 1469598103934665603ull;
+  auto mix = [&key](uint64_t v) {
+    key ^= v;
+    key *= 1099511628211ull;
+  };
+  mix(static_cast<uint64_t>(n_lists));
+  mix(static_cast<uint64_t>(n));
+  for (int l = 0; l < n_lists; ++l)
+    for (int t = 0; t < n; ++t)
+      mix(reinterpret_cast<uint64_t>(lists[l][t].data_ptr()));
+  for (int t = 0; t < n; ++t)
+    mix(static_cast<uint64_t>(lists[0][t].numel()));
+  {
+    std::lock_guard<std::mutex> lock(g_meta_mutex);
+    auto it = g_meta_cache.find(key);
+    if (it != g_meta_cache.end()) return it->second;
+  }
+
   auto cpu_i64 = torch::TensorOptions().dtype(torch::kLong);
   auto cpu_i32 = torch::TensorOptions().dtype(torch::kInt);
   auto ptrs_cpu = torch::empty({n_lists, n}, cpu_i64);
   auto* pp = ptrs_cpu.data_ptr<int64_t>();
+  auto sizes_cpu = torch::empty({n}, cpu_i64);
+  auto* sp = sizes_cpu.data_ptr<int64_t>();
   std::vector<int> ct, clo, chi;
   for (int t = 0; t < n; ++t) {
     const int64_t numel = lists[0][t].numel();
+    sp[t] = numel;
     for (int l = 0; l < n_lists; ++l) {
       TORCH_CHECK(lists[l][t].is_contiguous() &&
                       lists[l][t].scalar_type() == torch::kFloat32,
@@ -63,7 +102,14 @@ ChunkedList build_chunks(const std::vector<std::vector<torch::Tensor>>& lists,
     cp[i * 3 + 1] = clo[i];
     cp[i * 3 + 2] = chi[i];
   }
-  return {ptrs_cpu.to(device), chunks_cpu.to(device), n_chunks};
+  ChunkedList out{ptrs_cpu.to(device), chunks_cpu.to(device),
+                  sizes_cpu.to(device), n_chunks};
+  {
+    std::lock_guard<std::mutex> lock(g_meta_mutex);
+    if (g_meta_cache.size() > 64) g_meta_cache.clear();  // bound growth
+    g_meta_cache.emplace(key, out);
+  }
+  return out;
 }
 
 __device__ __forceinline__ int64_t chunk_off(const int* info) {
@@ -204,25 +250,15 @@ __global__ void adam_kernel(const int64_t* __restrict__ ptrs,
   }
 }
 
-static torch::Tensor sizes_tensor(const std::vector<torch::Tensor>& ts,
-                                  const torch::Device& device) {
-  auto cpu = torch::empty({static_cast<int64_t>(ts.size())},
-                          torch::TensorOptions().dtype(torch::kLong));
-  auto* sp = cpu.data_ptr<int64_t>();
-  for (size_t i = 0; i < ts.size(); ++i) sp[i] = ts[i].numel();
-  return cpu.to(device);
-}
-
 torch::Tensor multi_tensor_l2norm_sq(std::vector<torch::Tensor> tensors) {
   TORCH_CHECK(!tensors.empty(), "l2norm: empty list");
   auto device = tensors[0].device();
   auto meta = build_chunks({tensors}, device);
-  auto sizes = sizes_tensor(tensors, device);
   auto out = torch::zeros({1}, tensors[0].options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(l2norm_sq_kernel, dim3(meta.n_chunks), dim3(kThreads), 0,
                      stream, meta.ptrs.data_ptr<int64_t>(),
-                     sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(),
+                     meta.sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(),
                      out.data_ptr<float>(), meta.n_chunks);
   return out.squeeze(0);
 }
@@ -231,12 +267,11 @@ void multi_tensor_clip_scale(std::vector<torch::Tensor> grads,
                              torch::Tensor gnorm_sq, double max_norm) {
   auto device = grads[0].device();
   auto meta = build_chunks({grads}, device);
-  auto sizes = sizes_tensor(grads, device);
   auto gsq = gnorm_sq.reshape({1}).contiguous();
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(clip_scale_kernel, dim3(meta.n_chunks), dim3(kThreads), 0,
                      stream, meta.ptrs.data_ptr<int64_t>(),
-                     sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(),
+                     meta.sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(),
                      gsq.data_ptr<float>(), static_cast<float>(max_norm),
                      meta.n_chunks);
 }
@@ -251,7 +286,6 @@ void fused_lamb(std::vector<torch::Tensor> params,
   TORCH_CHECK(n > 0, "fused_lamb: empty");
   auto device = params[0].device();
   auto meta = build_chunks({params, grads, ms, vs}, device);
-  auto sizes = sizes_tensor(params, device);
   auto norms = torch::zeros({n, 2}, params[0].options().dtype(torch::kFloat32));
   auto gsq = gnorm_sq.reshape({1}).contiguous();
   const float bc1 = bias_correction ? 1.f - powf(beta1, step) : 1.f;
@@ -260,7 +294,7 @@ void fused_lamb(std::vector<torch::Tensor> params,
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(lamb_stage1_kernel, dim3(meta.n_chunks), dim3(kThreads),
                      0, stream, meta.ptrs.data_ptr<int64_t>(),
-                     sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(),
+                     meta.sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(),
                      gsq.data_ptr<float>(), norms.data_ptr<float>(), n,
                      meta.n_chunks, static_cast<float>(beta1),
                      static_cast<float>(beta2), static_cast<float>(eps),
@@ -268,7 +302,7 @@ void fused_lamb(std::vector<torch::Tensor> params,
                      static_cast<float>(max_grad_norm));
   hipLaunchKernelGGL(lamb_stage2_kernel, dim3(meta.n_chunks), dim3(kThreads),
                      0, stream, meta.ptrs.data_ptr<int64_t>(),
-                     sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(),
+                     meta.sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(),
                      norms.data_ptr<float>(), n, meta.n_chunks,
                      static_cast<float>(lr), use_ratio);
 }
@@ -282,13 +316,12 @@ void fused_adam(std::vector<torch::Tensor> params,
   TORCH_CHECK(n > 0, "fused_adam: empty");
   auto device = params[0].device();
   auto meta = build_chunks({params, grads, ms, vs}, device);
-  auto sizes = sizes_tensor(params, device);
   const float bc1 = bias_correction ? 1.f - powf(beta1, step) : 1.f;
   const float bc2 = bias_correction ? 1.f - powf(beta2, step) : 1.f;
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(adam_kernel, dim3(meta.n_chunks), dim3(kThreads), 0,
                      stream, meta.ptrs.data_ptr<int64_t>(),
-                     sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(), n,
+                     meta.sizes.data_ptr<int64_t>(), meta.chunks.data_ptr<int>(), n,
                      meta.n_chunks, static_cast<float>(lr),
                      static_cast<float>(beta1), static_cast<float>(beta2),
                      static_cast<float>(eps), static_cast<float>(wd), bc1, bc2,
