@@ -39,10 +39,15 @@ from bert_pytorch_amd.optim import FusedLAMB, PolyWarmUpScheduler  # noqa: E402
 from bert_pytorch_amd.parallel import comm  # noqa: E402
 
 PHASES = {
+    # BERT-Large two-phase LAMB (BASELINE configs 2-3)
     1: dict(seq_len=128, local_batch=96, max_pred=20, lr=6e-3,
             named_global=65536),
     2: dict(seq_len=512, local_batch=16, max_pred=80, lr=4e-3,
             named_global=32768),
+    # RoBERTa-Large single-phase fused-Adam (BASELINE config 4):
+    # seq 512, local 16, global 8192, lr 4e-4 linear decay
+    3: dict(seq_len=512, local_batch=16, max_pred=80, lr=4e-4,
+            named_global=8192),
 }
 
 
@@ -51,7 +56,15 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=32)
     p.add_argument("--warmup", type=int, default=8)
-    p.add_argument("--phase", type=int, default=1, choices=[1, 2])
+    p.add_argument("--phase", type=int, default=1, choices=[1, 2, 3],
+                   help="1/2 = BERT-Large seq128/seq512 LAMB; "
+                        "3 = RoBERTa-Large seq512 fused-Adam (no NSP)")
+    p.add_argument("--optimizer", type=str, default=None,
+                   choices=["lamb", "adam"],
+                   help="default: lamb (phases 1-2), adam (phase 3)")
+    p.add_argument("--kfac", action="store_true",
+                   help="K-FAC preconditioner before the optimizer step "
+                        "(BASELINE config 5)")
     p.add_argument("--accumulation", type=int, default=8,
                    help="optimizer cadence in micro-steps; 0 = the named "
                         "config's full accumulation (update may not fire "
@@ -119,6 +132,10 @@ def main():
     if accum <= 0:
         accum = max(1, -(-phase["named_global"] // (world * bsz)))
 
+    opt_name = args.optimizer or ("adam" if args.phase == 3 else "lamb")
+    if args.phase == 3 and args.model_config.endswith(
+            "bert_large_uncased_config.json"):
+        args.model_config = "config/roberta_large_cased_config.json"
     cfg_path = args.model_config
     if not os.path.exists(cfg_path):
         # resolve relative to this file so bench.py works from any cwd
@@ -144,17 +161,33 @@ def main():
                           grad_compress=args.grad_compress)
     named = list(model.named_parameters())
     no_decay = ("bias", "LayerNorm", "qkv_bias")
-    optimizer = FusedLAMB(
-        [
-            {"params": [p for n, p in named if not any(d in n for d in no_decay)],
-             "weight_decay": 0.01},
-            {"params": [p for n, p in named if any(d in n for d in no_decay)],
-             "weight_decay": 0.0},
-        ],
-        lr=phase["lr"],
-        master_weights=args.pure_bf16 or args.bf16_weights,
-    )
+    groups = [
+        {"params": [p for n, p in named if not any(d in n for d in no_decay)],
+         "weight_decay": 0.01},
+        {"params": [p for n, p in named if any(d in n for d in no_decay)],
+         "weight_decay": 0.0},
+    ]
+    if opt_name == "adam":
+        from bert_pytorch_amd.optim import FusedAdam  # noqa: PLC0415
+
+        optimizer = FusedAdam(
+            groups, lr=phase["lr"],
+            master_weights=args.pure_bf16 or args.bf16_weights,
+        )
+    else:
+        optimizer = FusedLAMB(
+            groups, lr=phase["lr"],
+            master_weights=args.pure_bf16 or args.bf16_weights,
+        )
     scheduler = PolyWarmUpScheduler(optimizer, warmup=0.2843, total_steps=7038)
+    preconditioner = None
+    if args.kfac:
+        from bert_pytorch_amd.optim.kfac import KFAC  # noqa: PLC0415
+
+        preconditioner = KFAC(
+            model, optimizer, inv_update_interval=10,
+            skip_layers=["BertLMPredictionHead", "embedding"],
+        )
     gen = torch.Generator(device=device).manual_seed(args.seed + rank)
     # BERT uncased real vocab (30522) unless the model config is smaller
     vocab_unpadded = min(30522, config.vocab_size)
@@ -206,6 +239,8 @@ def main():
                 loss.backward()
         if sync:
             scheduler.step()
+            if preconditioner is not None:
+                preconditioner.step()
             optimizer.step()
             optimizer.zero_grad(set_to_none=True)
 
@@ -258,9 +293,11 @@ def main():
                 "seq_len": seq,
                 "local_batch": bsz,
                 "accumulation": accum,
-                "optimizer": "FusedLAMB (HIP multi-tensor)"
+                "optimizer": ("FusedAdam" if opt_name == "adam"
+                              else "FusedLAMB") + " (HIP multi-tensor)"
                 + (" + fp32 masters"
-                   if args.pure_bf16 or args.bf16_weights else ""),
+                   if args.pure_bf16 or args.bf16_weights else "")
+                + (" + K-FAC" if args.kfac else ""),
                 "weights": "bf16 (fp32 LAMB masters)" if args.pure_bf16
                 else ("bf16 matmul weights + fp32 LN/bias (fp32 LAMB "
                       "masters)" if args.bf16_weights
