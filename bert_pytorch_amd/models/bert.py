@@ -60,7 +60,7 @@ class LinearActivation(nn.Module):
         # activation names per the reference ACT2FN (src/modeling.py:139):
         # gelu / bias_gelu are the fused HIP path; tanh/relu/swish eager.
         if self.act in ("gelu", "bias_gelu"):
-            y = F.linear(x, self.weight, None)
+            y = ops.linear_nobias(x, self.weight)
             return ops.fused_bias_gelu(y, self.bias)
         y = F.linear(x, self.weight, self.bias)
         if self.act in ("tanh", "bias_tanh"):
@@ -170,7 +170,7 @@ class BertSelfOutput(nn.Module):
         self.dropout_prob = config.hidden_dropout_prob
 
     def forward(self, hidden: torch.Tensor, residual: torch.Tensor) -> torch.Tensor:
-        y = F.linear(hidden, self.dense.weight, None)  # bias folded into fusion
+        y = ops.linear_nobias(hidden, self.dense.weight)  # bias in fusion
         return ops.fused_bias_dropout_residual_ln(
             y, self.dense.bias, residual,
             self.LayerNorm.weight, self.LayerNorm.bias,
@@ -220,7 +220,7 @@ class BertOutput(nn.Module):
         self.dropout_prob = config.hidden_dropout_prob
 
     def forward(self, hidden: torch.Tensor, residual: torch.Tensor) -> torch.Tensor:
-        y = F.linear(hidden, self.dense.weight, None)
+        y = ops.linear_nobias(hidden, self.dense.weight)
         return ops.fused_bias_dropout_residual_ln(
             y, self.dense.bias, residual,
             self.LayerNorm.weight, self.LayerNorm.bias,

@@ -69,4 +69,5 @@ from .embedding import fused_embedding_ln_dropout  # noqa: E402,F401
 from .attention import fused_attention  # noqa: E402,F401
 from .cross_entropy import fused_cross_entropy  # noqa: E402,F401
 from .linear import fused_linear  # noqa: E402,F401
+from .matmul import linear_nobias  # noqa: E402,F401
 from .ffn import ffn_supported, fused_ffn  # noqa: E402,F401

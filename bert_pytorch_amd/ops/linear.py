@@ -31,10 +31,12 @@ class _FusedLinear(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
+        from .matmul import _wgrad  # noqa: PLC0415
+
         x2, weight = ctx.saved_tensors
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
         dx = dy2 @ weight
-        dw = dy2.t() @ x2
+        dw = _wgrad(dy2, x2.contiguous())
         db = extension().col_sum(dy2)
         if db.dtype != dy2.dtype:
             db = db.to(dy2.dtype)

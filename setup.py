@@ -22,6 +22,7 @@ sources = [
     "csrc/ops/embedding.hip",
     "csrc/ops/cross_entropy.hip",
     "csrc/ops/attention.hip",
+    "csrc/ops/wgrad.hip",
     "csrc/ops/gemm_epilogue.cpp",
     "csrc/optim/multi_tensor.hip",
     "csrc/tok/tokenizer.cpp",

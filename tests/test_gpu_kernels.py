@@ -660,3 +660,43 @@ def test_attention_dropout_numerics(B, S, NH):
         dout.bfloat16(), qkv, seqlens, out, lse, dmask, NH, p, 77, 5
     )
     assert rel_err(dqkv, qr.grad) < 6e-2
+
+
+# ---------------------------------------------------------------------------
+# split-K MFMA wgrad GEMM (csrc/ops/wgrad.hip)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("K,M,N", [
+    (12288, 3072, 1024),   # QKV wgrad, phase 1
+    (12288, 1024, 4096),   # FFN2 wgrad
+    (8192, 4096, 1024),    # FFN1 wgrad, phase 2
+    (8192, 1024, 1024),    # attn-out wgrad (splitk 8)
+    (4100, 1024, 1024),    # K not a multiple of 64: zero-pad tail path
+    (384, 128, 128),       # minimal tile
+])
+def test_wgrad_tn_parity(K, M, N):
+    torch.manual_seed(K + M + N)
+    dy = torch.randn(K, M, device=DEV, dtype=torch.bfloat16)
+    x = torch.randn(K, N, device=DEV, dtype=torch.bfloat16)
+    assert ext().wgrad_tn_supported(K, M, N)
+    out = ext().wgrad_tn(dy, x)
+    ref_fp32 = dy.float().t() @ x.float()
+    # fp32 accumulation inside the kernel; only the final bf16 cast and
+    # the split-K summation order differ from the fp32 reference
+    assert rel_err(out, ref_fp32) < 1e-2, rel_err(out, ref_fp32)
+
+
+def test_wgrad_routes_in_linear_nobias():
+    """linear_nobias backward produces the same dW as eager F.linear."""
+    torch.manual_seed(0)
+    x = torch.randn(4096, 1024, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(1024, 1024, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = ops.linear_nobias(x, w)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    torch.nn.functional.linear(x2, w2).backward(gy)
+    assert rel_err(w.grad, w2.grad.float()) < 1e-2
+    assert rel_err(x.grad, x2.grad.float()) < 1e-2
