@@ -46,7 +46,8 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor qkv,
                                          int64_t num_heads, double p,
                                          int64_t seed, int64_t offset);
 torch::Tensor tr16_probe(int64_t addr_mode);
-torch::Tensor wgrad_tn(torch::Tensor dy, torch::Tensor x);
+torch::Tensor wgrad_tn(torch::Tensor dy, torch::Tensor x,
+                       int64_t splitk_override);
 bool wgrad_tn_supported(int64_t K, int64_t M, int64_t N);
 bool wgrad_tn_profitable(int64_t K, int64_t M, int64_t N);
 torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
@@ -105,7 +106,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_fwd", &bpa::attention_fwd);
   m.def("attention_bwd", &bpa::attention_bwd);
   m.def("tr16_probe", &bpa::tr16_probe);
-  m.def("wgrad_tn", &bpa::wgrad_tn);
+  m.def("wgrad_tn", &bpa::wgrad_tn, py::arg("dy"), py::arg("x"),
+        py::arg("splitk_override") = 0);
   m.def("wgrad_tn_supported", &bpa::wgrad_tn_supported);
   m.def("wgrad_tn_profitable", &bpa::wgrad_tn_profitable);
   m.def("multi_tensor_l2norm_sq", &bpa::multi_tensor_l2norm_sq);

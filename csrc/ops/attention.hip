@@ -4,11 +4,14 @@
 // dropout, PV — src/modeling.py:403-429) with MFMA kernels built on
 // v_mfma_f32_16x16x32_bf16:
 //
-// Forward: grid (S/64 q-tiles, B*heads); 4 waves/block, 16 q-rows per
-// wave. K and V^T staged in LDS (+8 bf16 row pad, conflict-free b64
-// reads); online softmax with the swapped-QK^T trick (S^T = mfma(K, Q))
-// so each lane's P scores chain directly into the PV A-fragment; the
-// padding mask enters as per-sequence valid lengths. Dropout keep-masks
+// Forward: grid (S/128 q-tiles, B*heads); 4 waves/block, 2x16 q-rows
+// per wave. K and V staged NATURALLY in LDS (K at a 72-elem stride for
+// conflict-free b64 row reads; V at an 80-elem stride for conflict-free
+// ds_read_b64_tr_b16 hardware-transpose fragment reads - no transposed
+// image is ever built); online softmax with the swapped-QK^T trick
+// (S^T = mfma(K, Q)) so each lane's P scores chain directly into the
+// PV A-fragment; the padding mask enters as per-sequence valid
+// lengths. Dropout keep-masks
 // are pre-generated once per forward by dropout_mask_kernel (Philox
 // 4x32-10, one BIT per score, rows padded to whole 32-bit words) and
 // the [B*NH, S, ceil(S/32)] word tensor is read by forward and both

@@ -341,7 +341,8 @@ bool wgrad_tn_profitable(int64_t K, int64_t M, int64_t N) {
 }
 
 // dW = dy^T @ x; dy [K,M] bf16 row-major contiguous, x [K,N] likewise.
-torch::Tensor wgrad_tn(torch::Tensor dy, torch::Tensor x) {
+torch::Tensor wgrad_tn(torch::Tensor dy, torch::Tensor x,
+                       int64_t splitk_override) {
   TORCH_CHECK(dy.is_cuda() && dy.dim() == 2 && dy.is_contiguous() &&
                   dy.scalar_type() == torch::kBFloat16,
               "wgrad_tn: dy must be contiguous 2D bf16");
@@ -352,15 +353,17 @@ torch::Tensor wgrad_tn(torch::Tensor dy, torch::Tensor x) {
   TORCH_CHECK(x.size(0) == K, "wgrad_tn: K mismatch");
   TORCH_CHECK(wgrad_tn_supported(K, M, N), "wgrad_tn: unsupported shape");
   const bool big = false;  // 256-tile variant measured SLOWER (1 block/CU cannot hide latency); kept for reference
-  // split K: ~1 block/CU for the 256-tile (1 block/CU is its ceiling:
-  // 8 waves at >176 VGPRs), ~4 blocks/CU for the 128-tile
   const int bm = big ? 256 : wg::kBM, bn = big ? 256 : wg::kBN;
   const int tiles = (M / bm) * (N / bn);
-  const int target = big ? 320 : 1024;
+  // split-K sweep on MI355X (benchmarks/wgrad_sweep.py): best wall time
+  // lands at tiles*splitk ~ 512-768 with splitk <= 8 (beyond that the
+  // fp32 slab traffic of the combine outweighs the occupancy gain):
+  // QKV 127->99us (splitk 4), attn-out 56->42 (8), FFN1 ~tie (3)
   int splitk = 1;
-  while (tiles * (splitk + 1) <= target &&
+  while (splitk < 8 && tiles * (splitk + 1) <= 768 &&
          K / (splitk + 1) >= 8 * wg::kBK)
     ++splitk;
+  if (splitk_override > 0) splitk = static_cast<int>(splitk_override);
   int k_slice =
       ((K + splitk - 1) / splitk + wg::kBK - 1) / wg::kBK * wg::kBK;
   splitk = (K + k_slice - 1) / k_slice;  // drop empty tail slices
