@@ -230,6 +230,7 @@ def main():
             scores, rel, glabels = model(
                 ids, tt, mask, masked_lm_labels=labels,
                 max_predictions_per_seq=phase["max_pred"],
+                compute_mlm_loss=True,
             )
             loss = criterion(scores, rel, glabels, nsp) / accum
         if sync or not isinstance(model, torch.nn.parallel.DistributedDataParallel):

@@ -501,6 +501,7 @@ class BertForPreTraining(BertPreTrainedModel):
         attention_mask=None,
         masked_lm_labels=None,
         max_predictions_per_seq=None,
+        compute_mlm_loss=False,
     ):
         encoded, pooled = self.bert(input_ids, token_type_ids, attention_mask)
         sequence_output = encoded[-1] if isinstance(encoded, list) else encoded
@@ -540,7 +541,20 @@ class BertForPreTraining(BertPreTrainedModel):
             gathered_labels = flat_labels.index_select(0, positions)
         hidden = sequence_output.reshape(-1, sequence_output.shape[-1])
         masked_hidden = hidden.index_select(0, positions)
-        scores = self.cls.predictions(masked_hidden)
+        if compute_mlm_loss:
+            # fused decoder-GEMM + bias + CE path (ops/mlm.py): the
+            # first return element is the scalar MLM loss instead of
+            # the [P, V] scores; BertPretrainingCriterion detects the
+            # 0-dim tensor and adds only the NSP term
+            head = self.cls.predictions
+            scores = ops.mlm_decoder_loss(
+                head.transform(masked_hidden),
+                head.decoder.weight,
+                head.bias,
+                gathered_labels,
+            )
+        else:
+            scores = self.cls.predictions(masked_hidden)
         seq_rel = (
             self.cls.seq_relationship(pooled)
             if self.cls.seq_relationship is not None and pooled is not None

@@ -28,13 +28,20 @@ class BertPretrainingCriterion(torch.nn.Module):
         masked_lm_labels: torch.Tensor,
         next_sentence_labels: Optional[torch.Tensor],
     ) -> torch.Tensor:
-        if prediction_scores.dim() == 3:  # full-scores API (reference layout)
-            scores = prediction_scores.view(-1, self.vocab_size)
-            labels = masked_lm_labels.view(-1)
+        if prediction_scores.dim() == 0:
+            # model already fused the MLM head into the loss
+            # (BertForPreTraining.forward(compute_mlm_loss=True))
+            loss = prediction_scores
+        elif prediction_scores.dim() == 3:  # full-scores API (reference)
+            loss = ops.fused_cross_entropy(
+                prediction_scores.view(-1, self.vocab_size),
+                masked_lm_labels.view(-1),
+                ignore_index=-1,
+            )
         else:  # gathered masked rows
-            scores = prediction_scores
-            labels = masked_lm_labels
-        loss = ops.fused_cross_entropy(scores, labels, ignore_index=-1)
+            loss = ops.fused_cross_entropy(
+                prediction_scores, masked_lm_labels, ignore_index=-1
+            )
         if seq_relationship_score is not None and next_sentence_labels is not None:
             loss = loss + F.cross_entropy(
                 seq_relationship_score.view(-1, 2).float(),

@@ -71,3 +71,4 @@ from .cross_entropy import fused_cross_entropy  # noqa: E402,F401
 from .linear import fused_linear  # noqa: E402,F401
 from .matmul import linear_nobias  # noqa: E402,F401
 from .ffn import ffn_supported, fused_ffn  # noqa: E402,F401
+from .mlm import mlm_decoder_loss  # noqa: E402,F401

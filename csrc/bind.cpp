@@ -48,6 +48,11 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor qkv,
 torch::Tensor tr16_probe(int64_t addr_mode);
 torch::Tensor wgrad_tn(torch::Tensor dy, torch::Tensor x,
                        int64_t splitk_override);
+std::vector<torch::Tensor> mlm_head_fwd(torch::Tensor h, torch::Tensor w,
+                                        torch::Tensor bias,
+                                        torch::Tensor labels,
+                                        int64_t ignore_index);
+bool mlm_head_supported(int64_t P, int64_t V, int64_t K);
 bool wgrad_tn_supported(int64_t K, int64_t M, int64_t N);
 bool wgrad_tn_profitable(int64_t K, int64_t M, int64_t N);
 torch::Tensor attention_bwd(torch::Tensor dout, torch::Tensor qkv,
@@ -109,6 +114,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_tn", &bpa::wgrad_tn, py::arg("dy"), py::arg("x"),
         py::arg("splitk_override") = 0);
   m.def("wgrad_tn_supported", &bpa::wgrad_tn_supported);
+  m.def("mlm_head_fwd", &bpa::mlm_head_fwd,
+        "MFMA MLM-decoder GEMM with fused bias + CE-forward epilogue");
+  m.def("mlm_head_supported", &bpa::mlm_head_supported);
   m.def("wgrad_tn_profitable", &bpa::wgrad_tn_profitable);
   m.def("multi_tensor_l2norm_sq", &bpa::multi_tensor_l2norm_sq);
   m.def("multi_tensor_clip_scale", &bpa::multi_tensor_clip_scale);

@@ -1,0 +1,325 @@
+// Hand-written MFMA MLM-decoder GEMM with fused bias + cross-entropy
+// forward for MI355X (gfx950).
+//
+// logits[P,V] = h[P,K] @ W[V,K]^T + bias[V], V = vocab (~30528),
+// K = hidden (1024), P = gathered masked rows (B * max_preds) — the
+// single biggest GEMM of the reference's MLM head
+// (src/modeling.py:570-578 decoder matmul + run_pretraining.py:58-72
+// CrossEntropyLoss). One kernel computes the GEMM, adds the bias, and
+// produces the cross-entropy forward statistics in the epilogue:
+// per-(row, column-tile) online (max, sum-exp) fp32 partials folded by
+// a tiny second kernel into the per-row logsumexp + NLL loss. The
+// separate full [P,V] read of a standalone CE-forward pass disappears;
+// backward reuses ce_bwd (csrc/ops/cross_entropy.hip) on the bf16
+// logits this kernel stores.
+//
+// Design notes (vs csrc/ops/wgrad.hip, the house split-K GEMM): here
+// BOTH operands are k-contiguous (h rows and W rows are K-major), so
+// MFMA fragments are two plain ds_read_b64 per fragment from
+// row-major [rows][kBK] LDS images — no transpose reads. The LDS row
+// stride is 36 elems (18 dwords): lane li's fragment base li*18 dw
+// mod 32 walks every even bank exactly once across the 16-lane group,
+// so both b64 fragment reads are bank-conflict-free. The output is
+// huge (P/128 * ceil(V/128) blocks = ~2400 workgroups at production
+// shapes), so no split-K is needed; K=1024 runs as 32 double-buffered
+// 32-deep T14 (issue-early/write-late) k-steps, one barrier per step.
+//
+// Numerics: fp32 MFMA accumulation; the CE statistics are computed
+// from the bf16-ROUNDED logits (exactly the values backward re-reads),
+// so forward lse and backward softmax see identical inputs, matching
+// the standalone ce_fwd/ce_bwd pair bit-for-bit in structure.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "../common.h"
+
+namespace bpa {
+
+namespace mh {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4v;
+
+#define MFMA16(A, B, C) \
+  __builtin_amdgcn_mfma_f32_16x16x32_bf16((A), (B), (C), 0, 0, 0)
+
+constexpr int kBM = 128;     // rows (gathered masked positions)
+constexpr int kBN = 128;     // vocab columns
+constexpr int kBK = 32;      // K step (one MFMA depth)
+constexpr int kStride = 36;  // LDS row stride, elems (18 dw: b64-clean)
+constexpr int kTile = kBM * kStride;  // one operand image, elems
+
+// A/B fragment from a k-contiguous row-major [rows][kBK] LDS image:
+// lane (g = lane>>4, li = lane&15) holds row rb+li, k = 4g..4g+3 and
+// 16+4g..16+4g+3 (the probe-verified gfx950 16x16x32 layout).
+__device__ __forceinline__ bf16x8 frag_k(const __bf16* img, int rb) {
+  const int lane = threadIdx.x & 63;
+  const int g = (lane >> 4) & 3, li = lane & 15;
+  const __bf16* p = img + (rb + li) * kStride + g * 4;
+  union {
+    bf16x8 v;
+    bf16x4v h[2];
+  } r;
+  r.h[0] = *reinterpret_cast<const bf16x4v*>(p);
+  r.h[1] = *reinterpret_cast<const bf16x4v*>(p + 16);
+  return r.v;
+}
+
+}  // namespace mh
+
+using mh::bf16x8;
+using mh::f32x4;
+
+// grid: (P/128, ceil(V/128)); block 256 (4 waves as 2x2, 64x64 each).
+__global__ __launch_bounds__(256) void mlm_fwd_kernel(
+    const __bf16* __restrict__ h,     // [P, K]
+    const __bf16* __restrict__ w,     // [V, K]
+    const float* __restrict__ bias,   // [V]
+    __bf16* __restrict__ logits,      // [P, V]
+    float* __restrict__ part,         // [P, nTiles, 2] (max, sumexp)
+    int P, int V, int K) {
+  const int m0 = blockIdx.x * mh::kBM;
+  const int n0 = blockIdx.y * mh::kBN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int wi = wave >> 1, wj = wave & 1;  // 2x2 wave grid, 64x64 each
+  const int g = (lane >> 4), li = lane & 15;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* lds = reinterpret_cast<__bf16*>(smem);
+
+  // staging map: 2 threads per row, each writes two uint4 (16 elems)
+  const int st_r = tid >> 1;             // 0..127
+  const int st_c = (tid & 1) * 16;       // 0 or 16 (of kBK=32)
+
+  const int64_t h_row = static_cast<int64_t>(m0 + st_r) * K;
+  // vocab tail tile: clamp W row (always-legal load; epilogue masks)
+  const int64_t w_row = static_cast<int64_t>(min(n0 + st_r, V - 1)) * K;
+
+  uint4 pf_h[2], pf_w[2];
+  auto issue_loads = [&](int k0) {
+    const uint4* hs = reinterpret_cast<const uint4*>(h + h_row + k0 + st_c);
+    const uint4* ws = reinterpret_cast<const uint4*>(w + w_row + k0 + st_c);
+    pf_h[0] = hs[0];
+    pf_h[1] = hs[1];
+    pf_w[0] = ws[0];
+    pf_w[1] = ws[1];
+  };
+
+  f32x4 acc[4][4] = {};
+  issue_loads(0);
+
+  int buf = 0;
+  for (int k0 = 0; k0 < K; k0 += mh::kBK) {
+    const int boff = buf * 2 * mh::kTile;
+    // b64 staging writes: the 72-B row stride is 8-byte aligned (not
+    // 16), so uint4 stores would fault on odd rows; uint2 pairs keep
+    // the same bytes with dw offsets row*18 + half*8 + {0,2,4,6} —
+    // distinct banks across each 8-lane group (conflict-free)
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      const uint2* hh = reinterpret_cast<const uint2*>(&pf_h[q]);
+      const uint2* wh = reinterpret_cast<const uint2*>(&pf_w[q]);
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        *reinterpret_cast<uint2*>(
+            &lds[boff + st_r * mh::kStride + st_c + 8 * q + 4 * u]) = hh[u];
+        *reinterpret_cast<uint2*>(
+            &lds[boff + mh::kTile + st_r * mh::kStride + st_c + 8 * q +
+                 4 * u]) = wh[u];
+      }
+    }
+    __syncthreads();
+    issue_loads(k0 + mh::kBK < K ? k0 + mh::kBK : k0);  // T14 issue-early
+
+    const __bf16* ht = lds + boff;
+    const __bf16* wt = lds + boff + mh::kTile;
+    {
+      bf16x8 af[4], bfr[4];
+#pragma unroll
+      for (int t = 0; t < 4; ++t) af[t] = mh::frag_k(ht, wi * 64 + t * 16);
+#pragma unroll
+      for (int t = 0; t < 4; ++t) bfr[t] = mh::frag_k(wt, wj * 64 + t * 16);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ti = 0; ti < 4; ++ti)
+#pragma unroll
+        for (int tj = 0; tj < 4; ++tj)
+          acc[ti][tj] = MFMA16(af[ti], bfr[tj], acc[ti][tj]);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    buf ^= 1;
+  }
+
+  // ---- epilogue: bias add, bf16 logits store, CE partial stats ----
+  float bv[4];
+  bool cv[4];  // column-valid (vocab tail tile)
+#pragma unroll
+  for (int tj = 0; tj < 4; ++tj) {
+    const int nj = n0 + wj * 64 + tj * 16 + li;
+    cv[tj] = nj < V;
+    bv[tj] = cv[tj] ? bias[nj] : 0.f;
+  }
+
+  // per-lane per-row (4 cols) online stats, then butterfly over the
+  // 16-lane li-group so every lane of the group holds the row stat
+  __syncthreads();  // all waves done with the last LDS buffer
+  float* stats = reinterpret_cast<float*>(smem);  // [2(wi)][64 rows][2(wj)][2]
+#pragma unroll
+  for (int ti = 0; ti < 4; ++ti) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int mi = m0 + wi * 64 + ti * 16 + g * 4 + r;
+      float m = -INFINITY, s = 0.f;
+#pragma unroll
+      for (int tj = 0; tj < 4; ++tj) {
+        const int nj = n0 + wj * 64 + tj * 16 + li;
+        const __bf16 ob = __bf16(acc[ti][tj][r] + bv[tj]);
+        if (cv[tj]) {
+          logits[static_cast<int64_t>(mi) * V + nj] = ob;
+          const float f = static_cast<float>(ob);
+          if (f > m) {
+            s *= __expf(m - f);
+            m = f;
+          }
+          s += __expf(f - m);
+        }
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) {
+        const float m2 = __shfl_xor(m, off, 64);
+        const float s2 = __shfl_xor(s, off, 64);
+        const float mn = fmaxf(m, m2);
+        s = (s == 0.f ? 0.f : s * __expf(m - mn)) +
+            (s2 == 0.f ? 0.f : s2 * __expf(m2 - mn));
+        m = mn;
+      }
+      if (li == 0) {
+        const int row = wi * 64 + ti * 16 + g * 4 + r;
+        stats[(row * 2 + wj) * 2 + 0] = m;
+        stats[(row * 2 + wj) * 2 + 1] = s;
+      }
+    }
+  }
+  __syncthreads();
+  // fold the two wj half-tiles and publish the block's partial
+  if (tid < mh::kBM) {
+    float m1 = stats[(tid * 2 + 0) * 2 + 0], s1 = stats[(tid * 2 + 0) * 2 + 1];
+    float m2 = stats[(tid * 2 + 1) * 2 + 0], s2 = stats[(tid * 2 + 1) * 2 + 1];
+    const float mn = fmaxf(m1, m2);
+    const float sn = (s1 == 0.f ? 0.f : s1 * __expf(m1 - mn)) +
+                     (s2 == 0.f ? 0.f : s2 * __expf(m2 - mn));
+    float* p = part +
+               (static_cast<int64_t>(m0 + tid) * gridDim.y + blockIdx.y) * 2;
+    p[0] = mn;
+    p[1] = sn;
+  }
+}
+
+// fold per-row partials -> lse, loss (mean over non-ignored rows done
+// host-side from loss_sum / count, matching ce_fwd's contract)
+__global__ void mlm_fold_kernel(const float* __restrict__ part,
+                                const __bf16* __restrict__ logits,
+                                const int64_t* __restrict__ labels,
+                                float* __restrict__ loss_sum,
+                                int* __restrict__ count,
+                                float* __restrict__ lse_out, int nTiles,
+                                int V, int64_t ignore_index) {
+  const int row = blockIdx.x;
+  const float* pr = part + static_cast<int64_t>(row) * nTiles * 2;
+  float m = -INFINITY, s = 0.f;
+  for (int t = threadIdx.x; t < nTiles; t += blockDim.x) {
+    const float m2 = pr[t * 2], s2 = pr[t * 2 + 1];
+    const float mn = fmaxf(m, m2);
+    s = (s == 0.f ? 0.f : s * __expf(m - mn)) +
+        (s2 == 0.f ? 0.f : s2 * __expf(m2 - mn));
+    m = mn;
+  }
+  __shared__ float sm[256], ss[256];
+  sm[threadIdx.x] = m;
+  ss[threadIdx.x] = s;
+  __syncthreads();
+  for (int stride = blockDim.x / 2; stride > 0; stride >>= 1) {
+    if (threadIdx.x < stride) {
+      const float m2 = sm[threadIdx.x + stride], s2 = ss[threadIdx.x + stride];
+      const float m1 = sm[threadIdx.x], s1 = ss[threadIdx.x];
+      const float mn = fmaxf(m1, m2);
+      sm[threadIdx.x] = mn;
+      ss[threadIdx.x] = (s1 == 0.f ? 0.f : s1 * __expf(m1 - mn)) +
+                        (s2 == 0.f ? 0.f : s2 * __expf(m2 - mn));
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    const float lse = sm[0] + __logf(ss[0]);
+    lse_out[row] = lse;
+    const int64_t label = labels[row];
+    if (label != ignore_index) {
+      const float xl = static_cast<float>(
+          logits[static_cast<int64_t>(row) * V + label]);
+      atomicAdd(loss_sum, lse - xl);
+      atomicAdd(count, 1);
+    }
+  }
+}
+
+bool mlm_head_supported(int64_t P, int64_t V, int64_t K) {
+  return P % mh::kBM == 0 && K % mh::kBK == 0 && K >= 64 && V >= 2 &&
+         K % 8 == 0;
+}
+
+// h [P,K] bf16, w [V,K] bf16, bias [V] fp32, labels [P] int64 ->
+// {logits [P,V] bf16, loss_sum f32, count f32, lse [P] f32}
+std::vector<torch::Tensor> mlm_head_fwd(torch::Tensor h, torch::Tensor w,
+                                        torch::Tensor bias,
+                                        torch::Tensor labels,
+                                        int64_t ignore_index) {
+  TORCH_CHECK(h.is_cuda() && h.dim() == 2 && h.is_contiguous() &&
+                  h.scalar_type() == torch::kBFloat16,
+              "mlm_head_fwd: h must be contiguous 2D bf16");
+  TORCH_CHECK(w.is_cuda() && w.dim() == 2 && w.is_contiguous() &&
+                  w.scalar_type() == torch::kBFloat16,
+              "mlm_head_fwd: w must be contiguous 2D bf16");
+  TORCH_CHECK(bias.is_cuda() && bias.dim() == 1 &&
+                  bias.scalar_type() == torch::kFloat32,
+              "mlm_head_fwd: bias must be 1D fp32");
+  const int P = h.size(0), K = h.size(1), V = w.size(0);
+  TORCH_CHECK(w.size(1) == K && bias.size(0) == V,
+              "mlm_head_fwd: shape mismatch");
+  TORCH_CHECK(mlm_head_supported(P, V, K), "mlm_head_fwd: unsupported shape");
+  auto labels_c = labels.contiguous();
+  TORCH_CHECK(labels_c.size(0) == P, "mlm_head_fwd: labels/P mismatch");
+
+  const int nTiles = (V + mh::kBN - 1) / mh::kBN;
+  auto fopts = h.options().dtype(torch::kFloat32);
+  auto logits = torch::empty({P, V}, h.options());
+  auto part = torch::empty({P, nTiles, 2}, fopts);
+  auto loss_sum = torch::zeros({1}, fopts);
+  auto count = torch::zeros({1}, h.options().dtype(torch::kInt32));
+  auto lse = torch::empty({P}, fopts);
+  auto stream = at::hip::getCurrentHIPStream();
+
+  const size_t lds = 4 * mh::kTile * sizeof(__bf16);
+  HIP_CHECK(hipFuncSetAttribute(
+      reinterpret_cast<const void*>(&mlm_fwd_kernel),
+      hipFuncAttributeMaxDynamicSharedMemorySize, lds));
+  hipLaunchKernelGGL(mlm_fwd_kernel, dim3(P / mh::kBM, nTiles), dim3(256),
+                     lds, stream,
+                     reinterpret_cast<const __bf16*>(h.data_ptr()),
+                     reinterpret_cast<const __bf16*>(w.data_ptr()),
+                     bias.data_ptr<float>(),
+                     reinterpret_cast<__bf16*>(logits.data_ptr()),
+                     part.data_ptr<float>(), P, V, K);
+  hipLaunchKernelGGL(mlm_fold_kernel, dim3(P), dim3(256), 0, stream,
+                     part.data_ptr<float>(),
+                     reinterpret_cast<const __bf16*>(logits.data_ptr()),
+                     labels_c.data_ptr<int64_t>(), loss_sum.data_ptr<float>(),
+                     count.data_ptr<int>(), lse.data_ptr<float>(), nTiles, V,
+                     ignore_index);
+  return {logits, loss_sum.squeeze(0), count.squeeze(0).to(torch::kFloat32),
+          lse};
+}
+
+}  // namespace bpa

@@ -335,6 +335,7 @@ def forward_backward_pass(model, criterion, scaler, batch, args, sync_grads,
         scores, seq_rel, gathered_labels = model(
             input_ids, segment_ids, input_mask, masked_lm_labels=mlm_labels,
             max_predictions_per_seq=args.max_predictions_per_seq,
+            compute_mlm_loss=True,
         )
         loss = criterion(scores, seq_rel, gathered_labels, nsp_labels)
         loss = loss / args.accumulation_steps

@@ -23,6 +23,7 @@ sources = [
     "csrc/ops/cross_entropy.hip",
     "csrc/ops/attention.hip",
     "csrc/ops/wgrad.hip",
+    "csrc/ops/mlm_head.hip",
     "csrc/ops/gemm_epilogue.cpp",
     "csrc/optim/multi_tensor.hip",
     "csrc/tok/tokenizer.cpp",

@@ -192,3 +192,37 @@ def test_padded_gather_matches_nonzero_gather(tiny_config):
             continue
         torch.testing.assert_close(p1.grad, p2.grad, rtol=1e-4, atol=1e-6,
                                    msg=lambda m, n=n1: f"{n}: {m}")
+
+
+def test_fused_mlm_loss_path_matches_scores_path(tiny_config):
+    """compute_mlm_loss=True returns the scalar MLM loss in place of the
+    scores and the criterion's 0-dim branch yields the same total loss
+    (on CPU the fused op runs its eager fp32 oracle)."""
+    model = BertForPreTraining(tiny_config).eval()
+    criterion = BertPretrainingCriterion(tiny_config.vocab_size)
+    ids, tt, mask, labels, nsp = _batch(tiny_config)
+    with torch.no_grad():
+        scores, rel, glabels = model(ids, tt, mask, masked_lm_labels=labels)
+        loss_scores = criterion(scores, rel, glabels, nsp)
+        fused, rel2, gl2 = model(
+            ids, tt, mask, masked_lm_labels=labels, compute_mlm_loss=True
+        )
+        assert fused.dim() == 0
+        loss_fused = criterion(fused, rel2, gl2, nsp)
+    torch.testing.assert_close(loss_scores, loss_fused, rtol=1e-5, atol=1e-5)
+
+
+def test_fused_mlm_loss_backward_cpu(tiny_config):
+    """Gradients flow through the fused-loss path (tied decoder weight,
+    bias, and the encoder) on the CPU oracle."""
+    model = BertForPreTraining(tiny_config)
+    criterion = BertPretrainingCriterion(tiny_config.vocab_size)
+    ids, tt, mask, labels, nsp = _batch(tiny_config)
+    loss, rel, gl = model(
+        ids, tt, mask, masked_lm_labels=labels, compute_mlm_loss=True,
+        max_predictions_per_seq=4,
+    )
+    criterion(loss, rel, gl, nsp).backward()
+    emb = model.bert.embeddings.word_embeddings.weight
+    assert emb.grad is not None and torch.isfinite(emb.grad).all()
+    assert model.cls.predictions.bias.grad is not None
