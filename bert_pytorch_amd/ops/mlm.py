@@ -22,6 +22,8 @@ is the right MI355X trade. See docs/KERNELS.md.
 
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn.functional as F
 
@@ -87,7 +89,11 @@ def mlm_decoder_loss(
             torch.is_autocast_enabled()
             and torch.get_autocast_gpu_dtype() == torch.bfloat16
         )
-        if bf16 and ext.mlm_head_supported(p_pad, V, K):
+        if (
+            bf16
+            and ext.mlm_head_supported(p_pad, V, K)
+            and os.environ.get("BPA_NO_FUSED_MLM") != "1"  # A/B toggle
+        ):
             h = hidden.to(torch.bfloat16)
             w = weight.to(torch.bfloat16)
             b = bias.float()
