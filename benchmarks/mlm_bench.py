@@ -9,7 +9,13 @@ K=1024):
 Run on a GPU box: python benchmarks/mlm_bench.py
 """
 
+import os
+import sys
 import time
+
+sys.path.insert(
+    0, os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+)
 
 import torch
 import torch.nn.functional as F
@@ -32,7 +38,9 @@ def timeit(fn, iters=50, warmup=10):
 
 def main():
     ext = ops.extension()
-    for P, V, K in [(1280, 30528, 1024), (2560, 30528, 1024)]:
+    # 1920 = the phase-1 bench micro-batch (local_batch 96 x 20 preds)
+    for P, V, K in [(1280, 30528, 1024), (1920, 30528, 1024),
+                    (2560, 30528, 1024)]:
         torch.manual_seed(0)
         h = (torch.randn(P, K, device=DEV) * 0.5).bfloat16()
         w = (torch.randn(V, K, device=DEV) * 0.05).bfloat16()
