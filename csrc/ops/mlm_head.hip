@@ -45,26 +45,41 @@ typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4v;
 #define MFMA16(A, B, C) \
   __builtin_amdgcn_mfma_f32_16x16x32_bf16((A), (B), (C), 0, 0, 0)
 
-constexpr int kBM = 128;     // rows (gathered masked positions)
-constexpr int kBN = 128;     // vocab columns
-constexpr int kBK = 32;      // K step (one MFMA depth)
-constexpr int kStride = 36;  // LDS row stride, elems (18 dw: b64-clean)
-constexpr int kTile = kBM * kStride;  // one operand image, elems
+constexpr int kBM = 128;  // rows (gathered masked positions)
+constexpr int kBN = 128;  // vocab columns
+constexpr int kBK = 64;   // K step (two MFMA depths per staged tile)
+constexpr int kTile = kBM * kBK;  // one operand image, elems (16 KB)
 
-// A/B fragment from a k-contiguous row-major [rows][kBK] LDS image:
-// lane (g = lane>>4, li = lane&15) holds row rb+li, k = 4g..4g+3 and
-// 16+4g..16+4g+3 (the probe-verified gfx950 16x16x32 layout).
-__device__ __forceinline__ bf16x8 frag_k(const __bf16* img, int rb) {
+// LDS images are LANE-LINEAR (global_load_lds writes wave-uniform base
+// + lane*16, so no padding is possible); a linear 128-B row would put
+// every fragment lane in the same bank span (the glds K-tile trap), so
+// the SOURCE address carries a 16-B-granule XOR swizzle instead:
+//   LDS[row r][granule p] = global[row r][granule p ^ (r & 7)]
+// A row is exactly one 128-B cache line, and the granule permutation
+// stays inside it, so global coalescing is untouched. Fragment b64
+// reads then spread the 16-lane li-group over 8 distinct granules
+// (2-way bank conflict — lanes li and li+8 share one), which is noise
+// next to the 32 MFMAs per k-step.
+__device__ __forceinline__ int swz_off(int r, int k) {
+  // element byte offset of (row r, k) inside one swizzled image
+  return r * (kBK * 2) + ((((k >> 3) ^ (r & 7)) << 4) | ((k & 7) << 1));
+}
+
+// A/B fragment for MFMA depth base ks (0 or 32): lane (g = lane>>4,
+// li = lane&15) holds row rb+li, k = ks+4g..+3 and ks+16+4g..+3 (the
+// probe-verified gfx950 16x16x32 layout).
+__device__ __forceinline__ bf16x8 frag_k(const char* img, int rb, int ks) {
   const int lane = threadIdx.x & 63;
   const int g = (lane >> 4) & 3, li = lane & 15;
-  const __bf16* p = img + (rb + li) * kStride + g * 4;
+  const int r = rb + li;
   union {
     bf16x8 v;
     bf16x4v h[2];
-  } r;
-  r.h[0] = *reinterpret_cast<const bf16x4v*>(p);
-  r.h[1] = *reinterpret_cast<const bf16x4v*>(p + 16);
-  return r.v;
+  } f;
+  f.h[0] = *reinterpret_cast<const bf16x4v*>(img + swz_off(r, ks + 4 * g));
+  f.h[1] =
+      *reinterpret_cast<const bf16x4v*>(img + swz_off(r, ks + 16 + 4 * g));
+  return f.v;
 }
 
 }  // namespace mh
@@ -88,60 +103,54 @@ __global__ __launch_bounds__(256) void mlm_fwd_kernel(
   const int g = (lane >> 4), li = lane & 15;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  __bf16* lds = reinterpret_cast<__bf16*>(smem);
+  constexpr int kTileB = mh::kTile * 2;  // one operand image, bytes
 
-  // staging map: 2 threads per row, each writes two uint4 (16 elems)
-  const int st_r = tid >> 1;             // 0..127
-  const int st_c = (tid & 1) * 16;       // 0 or 16 (of kBK=32)
-
-  const int64_t h_row = static_cast<int64_t>(m0 + st_r) * K;
-  // vocab tail tile: clamp W row (always-legal load; epilogue masks)
-  const int64_t w_row = static_cast<int64_t>(min(n0 + st_r, V - 1)) * K;
-
-  uint4 pf_h[2], pf_w[2];
-  auto issue_loads = [&](int k0) {
-    const uint4* hs = reinterpret_cast<const uint4*>(h + h_row + k0 + st_c);
-    const uint4* ws = reinterpret_cast<const uint4*>(w + w_row + k0 + st_c);
-    pf_h[0] = hs[0];
-    pf_h[1] = hs[1];
-    pf_w[0] = ws[0];
-    pf_w[1] = ws[1];
+  // global_load_lds staging: each of the 4 waves DMAs its 32-row slice
+  // of both operand images (4 calls x 8 rows x 128 B per matrix, 16 B
+  // per lane) straight from HBM to LDS — no staging VGPRs, no ds_write
+  // pass; the in-flight DMA overlaps the 32 MFMAs of the current
+  // k-step and the loop's single __syncthreads() drains it (vmcnt(0)
+  // inside the barrier's release — the simple verified glds shape).
+  const int st_sub = (lane >> 3);          // row within an 8-row piece
+  const int st_swz = ((lane & 7) ^ st_sub) * 8;  // swizzled source granule
+  auto stage = [&](int k0, int b) {
+    const char* base = smem + b * 2 * kTileB;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      const int r = wave * 32 + c * 8 + st_sub;
+      const __bf16* ga = h + static_cast<int64_t>(m0 + r) * K + k0 + st_swz;
+      // vocab tail tile: clamp W row (always-legal; epilogue masks)
+      const __bf16* gw =
+          w + static_cast<int64_t>(min(n0 + r, V - 1)) * K + k0 + st_swz;
+      const int roff = (wave * 32 + c * 8) * 128;  // wave-uniform
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)ga,
+          (__attribute__((address_space(3))) void*)(base + roff), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gw,
+          (__attribute__((address_space(3))) void*)(base + kTileB + roff),
+          16, 0, 0);
+    }
   };
 
   f32x4 acc[4][4] = {};
-  issue_loads(0);
+  stage(0, 0);
+  __syncthreads();
 
   int buf = 0;
   for (int k0 = 0; k0 < K; k0 += mh::kBK) {
-    const int boff = buf * 2 * mh::kTile;
-    // b64 staging writes: the 72-B row stride is 8-byte aligned (not
-    // 16), so uint4 stores would fault on odd rows; uint2 pairs keep
-    // the same bytes with dw offsets row*18 + half*8 + {0,2,4,6} —
-    // distinct banks across each 8-lane group (conflict-free)
+    if (k0 + mh::kBK < K) stage(k0 + mh::kBK, buf ^ 1);  // DMA under MFMA
+    const char* ht = smem + buf * 2 * kTileB;
+    const char* wt = ht + kTileB;
 #pragma unroll
-    for (int q = 0; q < 2; ++q) {
-      const uint2* hh = reinterpret_cast<const uint2*>(&pf_h[q]);
-      const uint2* wh = reinterpret_cast<const uint2*>(&pf_w[q]);
-#pragma unroll
-      for (int u = 0; u < 2; ++u) {
-        *reinterpret_cast<uint2*>(
-            &lds[boff + st_r * mh::kStride + st_c + 8 * q + 4 * u]) = hh[u];
-        *reinterpret_cast<uint2*>(
-            &lds[boff + mh::kTile + st_r * mh::kStride + st_c + 8 * q +
-                 4 * u]) = wh[u];
-      }
-    }
-    __syncthreads();
-    issue_loads(k0 + mh::kBK < K ? k0 + mh::kBK : k0);  // T14 issue-early
-
-    const __bf16* ht = lds + boff;
-    const __bf16* wt = lds + boff + mh::kTile;
-    {
+    for (int kd = 0; kd < 2; ++kd) {  // two MFMA depths per staged tile
       bf16x8 af[4], bfr[4];
 #pragma unroll
-      for (int t = 0; t < 4; ++t) af[t] = mh::frag_k(ht, wi * 64 + t * 16);
+      for (int t = 0; t < 4; ++t)
+        af[t] = mh::frag_k(ht, wi * 64 + t * 16, kd * 32);
 #pragma unroll
-      for (int t = 0; t < 4; ++t) bfr[t] = mh::frag_k(wt, wj * 64 + t * 16);
+      for (int t = 0; t < 4; ++t)
+        bfr[t] = mh::frag_k(wt, wj * 64 + t * 16, kd * 32);
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ti = 0; ti < 4; ++ti)
@@ -150,10 +159,15 @@ __global__ __launch_bounds__(256) void mlm_fwd_kernel(
           acc[ti][tj] = MFMA16(af[ti], bfr[tj], acc[ti][tj]);
       __builtin_amdgcn_s_setprio(0);
     }
+    __syncthreads();
     buf ^= 1;
   }
 
   // ---- epilogue: bias add, bf16 logits store, CE partial stats ----
+  // block-uniform full/tail-tile split: 238 of 239 production column
+  // tiles take the branch-free path (a per-element valid check makes
+  // hipcc branch + vmcnt(0) around every store)
+  const bool full = (n0 + mh::kBN) <= V;
   float bv[4];
   bool cv[4];  // column-valid (vocab tail tile)
 #pragma unroll
@@ -165,7 +179,6 @@ __global__ __launch_bounds__(256) void mlm_fwd_kernel(
 
   // per-lane per-row (4 cols) online stats, then butterfly over the
   // 16-lane li-group so every lane of the group holds the row stat
-  __syncthreads();  // all waves done with the last LDS buffer
   float* stats = reinterpret_cast<float*>(smem);  // [2(wi)][64 rows][2(wj)][2]
 #pragma unroll
   for (int ti = 0; ti < 4; ++ti) {
@@ -173,11 +186,11 @@ __global__ __launch_bounds__(256) void mlm_fwd_kernel(
     for (int r = 0; r < 4; ++r) {
       const int mi = m0 + wi * 64 + ti * 16 + g * 4 + r;
       float m = -INFINITY, s = 0.f;
+      if (full) {
 #pragma unroll
-      for (int tj = 0; tj < 4; ++tj) {
-        const int nj = n0 + wj * 64 + tj * 16 + li;
-        const __bf16 ob = __bf16(acc[ti][tj][r] + bv[tj]);
-        if (cv[tj]) {
+        for (int tj = 0; tj < 4; ++tj) {
+          const int nj = n0 + wj * 64 + tj * 16 + li;
+          const __bf16 ob = __bf16(acc[ti][tj][r] + bv[tj]);
           logits[static_cast<int64_t>(mi) * V + nj] = ob;
           const float f = static_cast<float>(ob);
           if (f > m) {
@@ -185,6 +198,21 @@ __global__ __launch_bounds__(256) void mlm_fwd_kernel(
             m = f;
           }
           s += __expf(f - m);
+        }
+      } else {
+#pragma unroll
+        for (int tj = 0; tj < 4; ++tj) {
+          const int nj = n0 + wj * 64 + tj * 16 + li;
+          const __bf16 ob = __bf16(acc[ti][tj][r] + bv[tj]);
+          if (cv[tj]) {
+            logits[static_cast<int64_t>(mi) * V + nj] = ob;
+            const float f = static_cast<float>(ob);
+            if (f > m) {
+              s *= __expf(m - f);
+              m = f;
+            }
+            s += __expf(f - m);
+          }
         }
       }
 #pragma unroll
