@@ -7,27 +7,36 @@
 // (src/modeling.py:570-578 decoder matmul + run_pretraining.py:58-72
 // CrossEntropyLoss). One kernel computes the GEMM, adds the bias, and
 // produces the cross-entropy forward statistics in the epilogue:
-// per-(row, column-tile) online (max, sum-exp) fp32 partials folded by
-// a tiny second kernel into the per-row logsumexp + NLL loss. The
+// per-(row, column-stripe) online (max, sum-exp) fp32 partials folded
+// by a tiny second kernel into the per-row logsumexp + NLL loss. The
 // separate full [P,V] read of a standalone CE-forward pass disappears;
 // backward reuses ce_bwd (csrc/ops/cross_entropy.hip) on the bf16
 // logits this kernel stores.
 //
-// Design notes (vs csrc/ops/wgrad.hip, the house split-K GEMM): here
-// BOTH operands are k-contiguous (h rows and W rows are K-major), so
-// MFMA fragments are two plain ds_read_b64 per fragment from
-// row-major [rows][kBK] LDS images — no transpose reads. The LDS row
-// stride is 36 elems (18 dwords): lane li's fragment base li*18 dw
-// mod 32 walks every even bank exactly once across the 16-lane group,
-// so both b64 fragment reads are bank-conflict-free. The output is
-// huge (P/128 * ceil(V/128) blocks = ~2400 workgroups at production
-// shapes), so no split-K is needed; K=1024 runs as 32 double-buffered
-// 32-deep T14 (issue-early/write-late) k-steps, one barrier per step.
+// Decomposition — this shape is "M tiny, N huge": a classic 128x128
+// tiling re-streams the 62.5-MB weight matrix once per row tile
+// (P=1280 -> 10x = 625 MB of HBM, an ~80-us floor that is exactly
+// where hipBLASLt lands). Instead each block owns a 64-column vocab
+// stripe for ALL of P: grid = V/64 (~477) blocks; a block loops over P
+// in 128-row chunks with the k-loop innermost. W is then read from HBM
+// EXACTLY ONCE (62.5 MB total), its per-block 128-KB stripe stays
+// L2-resident across the row chunks, and the small h operand
+// (2.6-5 MB) is re-read by every block but from L2, not HBM. This is
+// the M=256 projection-GEMM recipe of the CDNA4 guide applied at
+// M=P: total HBM traffic ~ W + h + logits + partials ~= 150-170 MB.
+//
+// Staging is global_load_lds (no staging VGPRs, no ds_write pass),
+// double-buffered, one __syncthreads per 64-deep k-step. LDS images
+// are lane-linear (glds writes wave-uniform base + lane*16), so the
+// SOURCE address carries a 16-B-granule XOR swizzle:
+//   LDS[row r][granule p] = global[row r][granule p ^ (r & 7)]
+// A row is one 128-B cache line and the permutation stays inside it,
+// so coalescing is untouched; fragment b64 reads spread each 16-lane
+// group over 8 granules (2-way conflict), noise under the MFMAs.
 //
 // Numerics: fp32 MFMA accumulation; the CE statistics are computed
 // from the bf16-ROUNDED logits (exactly the values backward re-reads),
-// so forward lse and backward softmax see identical inputs, matching
-// the standalone ce_fwd/ce_bwd pair bit-for-bit in structure.
+// so forward lse and backward softmax see identical inputs.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -45,27 +54,19 @@ typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4v;
 #define MFMA16(A, B, C) \
   __builtin_amdgcn_mfma_f32_16x16x32_bf16((A), (B), (C), 0, 0, 0)
 
-constexpr int kBM = 128;  // rows (gathered masked positions)
-constexpr int kBN = 128;  // vocab columns
+constexpr int kBM = 64;   // W-rows (vocab stripe) per block
+constexpr int kChunk = 128;  // h-rows per chunk of the in-block P loop
 constexpr int kBK = 64;   // K step (two MFMA depths per staged tile)
-constexpr int kTile = kBM * kBK;  // one operand image, elems (16 KB)
+constexpr int kATileB = kChunk * kBK * 2;  // h image bytes (16 KB)
+constexpr int kWTileB = kBM * kBK * 2;     // W image bytes (8 KB)
+constexpr int kBufB = kATileB + kWTileB;   // one double-buffer half
 
-// LDS images are LANE-LINEAR (global_load_lds writes wave-uniform base
-// + lane*16, so no padding is possible); a linear 128-B row would put
-// every fragment lane in the same bank span (the glds K-tile trap), so
-// the SOURCE address carries a 16-B-granule XOR swizzle instead:
-//   LDS[row r][granule p] = global[row r][granule p ^ (r & 7)]
-// A row is exactly one 128-B cache line, and the granule permutation
-// stays inside it, so global coalescing is untouched. Fragment b64
-// reads then spread the 16-lane li-group over 8 distinct granules
-// (2-way bank conflict — lanes li and li+8 share one), which is noise
-// next to the 32 MFMAs per k-step.
+// element byte offset of (row r, k) inside one swizzled image
 __device__ __forceinline__ int swz_off(int r, int k) {
-  // element byte offset of (row r, k) inside one swizzled image
   return r * (kBK * 2) + ((((k >> 3) ^ (r & 7)) << 4) | ((k & 7) << 1));
 }
 
-// A/B fragment for MFMA depth base ks (0 or 32): lane (g = lane>>4,
+// fragment for MFMA depth base ks (0 or 32): lane (g = lane>>4,
 // li = lane&15) holds row rb+li, k = ks+4g..+3 and ks+16+4g..+3 (the
 // probe-verified gfx950 16x16x32 layout).
 __device__ __forceinline__ bf16x8 frag_k(const char* img, int rb, int ks) {
@@ -87,162 +88,170 @@ __device__ __forceinline__ bf16x8 frag_k(const char* img, int rb, int ks) {
 using mh::bf16x8;
 using mh::f32x4;
 
-// grid: (P/128, ceil(V/128)); block 256 (4 waves as 2x2, 64x64 each).
+// grid: ceil(V/64) blocks of 256 threads (4 waves); each wave owns a
+// 32-row x 64-col sub-tile of the current 128-row chunk.
 __global__ __launch_bounds__(256) void mlm_fwd_kernel(
     const __bf16* __restrict__ h,     // [P, K]
     const __bf16* __restrict__ w,     // [V, K]
     const float* __restrict__ bias,   // [V]
     __bf16* __restrict__ logits,      // [P, V]
-    float* __restrict__ part,         // [P, nTiles, 2] (max, sumexp)
+    float* __restrict__ part,         // [P, gridDim.x, 2] (max, sumexp)
     int P, int V, int K) {
-  const int m0 = blockIdx.x * mh::kBM;
-  const int n0 = blockIdx.y * mh::kBN;
+  const int n0 = blockIdx.x * mh::kBM;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
-  const int wi = wave >> 1, wj = wave & 1;  // 2x2 wave grid, 64x64 each
   const int g = (lane >> 4), li = lane & 15;
+  const int nChunks = P / mh::kChunk;
+  const int kSteps = K / mh::kBK;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  constexpr int kTileB = mh::kTile * 2;  // one operand image, bytes
 
-  // global_load_lds staging: each of the 4 waves DMAs its 32-row slice
-  // of both operand images (4 calls x 8 rows x 128 B per matrix, 16 B
-  // per lane) straight from HBM to LDS — no staging VGPRs, no ds_write
-  // pass; the in-flight DMA overlaps the 32 MFMAs of the current
-  // k-step and the loop's single __syncthreads() drains it (vmcnt(0)
-  // inside the barrier's release — the simple verified glds shape).
-  const int st_sub = (lane >> 3);          // row within an 8-row piece
+  // column stripe is fixed for the whole block: bias once, in registers
+  const bool full = (n0 + mh::kBM) <= V;
+  float bv[4];
+  bool cv[4];
+#pragma unroll
+  for (int tj = 0; tj < 4; ++tj) {
+    const int nj = n0 + tj * 16 + li;
+    cv[tj] = nj < V;
+    bv[tj] = cv[tj] ? bias[nj] : 0.f;
+  }
+
+  // global_load_lds staging of one (h-chunk, W-stripe) k-slice pair:
+  // per wave 4 A calls + 2 W calls x 8 rows x 128 B, 16 B per lane.
+  const int st_sub = (lane >> 3);                // row within 8-row piece
   const int st_swz = ((lane & 7) ^ st_sub) * 8;  // swizzled source granule
-  auto stage = [&](int k0, int b) {
-    const char* base = smem + b * 2 * kTileB;
+  auto stage = [&](int mc, int k0, int b) {
+    char* base = smem + b * mh::kBufB;
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       const int r = wave * 32 + c * 8 + st_sub;
-      const __bf16* ga = h + static_cast<int64_t>(m0 + r) * K + k0 + st_swz;
-      // vocab tail tile: clamp W row (always-legal; epilogue masks)
-      const __bf16* gw =
-          w + static_cast<int64_t>(min(n0 + r, V - 1)) * K + k0 + st_swz;
-      const int roff = (wave * 32 + c * 8) * 128;  // wave-uniform
+      const __bf16* ga =
+          h + static_cast<int64_t>(mc * mh::kChunk + r) * K + k0 + st_swz;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)ga,
-          (__attribute__((address_space(3))) void*)(base + roff), 16, 0, 0);
+          (__attribute__((address_space(3))) void*)(base +
+                                                    (wave * 32 + c * 8) *
+                                                        128),
+          16, 0, 0);
+    }
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      const int r = wave * 16 + c * 8 + st_sub;
+      // vocab tail stripe: clamp W row (always-legal; epilogue masks)
+      const __bf16* gw =
+          w + static_cast<int64_t>(min(n0 + r, V - 1)) * K + k0 + st_swz;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)gw,
-          (__attribute__((address_space(3))) void*)(base + kTileB + roff),
+          (__attribute__((address_space(3))) void*)(base + mh::kATileB +
+                                                    (wave * 16 + c * 8) *
+                                                        128),
           16, 0, 0);
     }
   };
 
-  f32x4 acc[4][4] = {};
-  stage(0, 0);
+  f32x4 acc[2][4] = {};
+  stage(0, 0, 0);
   __syncthreads();
 
-  int buf = 0;
-  for (int k0 = 0; k0 < K; k0 += mh::kBK) {
-    if (k0 + mh::kBK < K) stage(k0 + mh::kBK, buf ^ 1);  // DMA under MFMA
-    const char* ht = smem + buf * 2 * kTileB;
-    const char* wt = ht + kTileB;
+  int buf = 0, kk = 0, mc = 0;
+  const int total = nChunks * kSteps;
+  for (int s = 0; s < total; ++s) {
+    // DMA the next k-slice (possibly of the next row chunk) under the
+    // MFMAs of this one
+    int kk2 = kk + 1, mc2 = mc;
+    if (kk2 == kSteps) {
+      kk2 = 0;
+      ++mc2;
+    }
+    if (mc2 < nChunks) stage(mc2, kk2 * mh::kBK, buf ^ 1);
+
+    const char* at = smem + buf * mh::kBufB;
+    const char* wt = at + mh::kATileB;
 #pragma unroll
     for (int kd = 0; kd < 2; ++kd) {  // two MFMA depths per staged tile
-      bf16x8 af[4], bfr[4];
+      bf16x8 af[2], bfr[4];
 #pragma unroll
-      for (int t = 0; t < 4; ++t)
-        af[t] = mh::frag_k(ht, wi * 64 + t * 16, kd * 32);
+      for (int t = 0; t < 2; ++t)
+        af[t] = mh::frag_k(at, wave * 32 + t * 16, kd * 32);
 #pragma unroll
-      for (int t = 0; t < 4; ++t)
-        bfr[t] = mh::frag_k(wt, wj * 64 + t * 16, kd * 32);
+      for (int t = 0; t < 4; ++t) bfr[t] = mh::frag_k(wt, t * 16, kd * 32);
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int ti = 0; ti < 4; ++ti)
+      for (int ti = 0; ti < 2; ++ti)
 #pragma unroll
         for (int tj = 0; tj < 4; ++tj)
           acc[ti][tj] = MFMA16(af[ti], bfr[tj], acc[ti][tj]);
       __builtin_amdgcn_s_setprio(0);
     }
+
+    if (++kk == kSteps) {
+      // ---- chunk epilogue: bias, bf16 logits, CE stats (no LDS) ----
+      const int m0 = mc * mh::kChunk;
+#pragma unroll
+      for (int ti = 0; ti < 2; ++ti) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int mi = m0 + wave * 32 + ti * 16 + g * 4 + r;
+          float mx = -INFINITY, sx = 0.f;
+          if (full) {
+#pragma unroll
+            for (int tj = 0; tj < 4; ++tj) {
+              const int nj = n0 + tj * 16 + li;
+              const __bf16 ob = __bf16(acc[ti][tj][r] + bv[tj]);
+              logits[static_cast<int64_t>(mi) * V + nj] = ob;
+              const float f = static_cast<float>(ob);
+              if (f > mx) {
+                sx *= __expf(mx - f);
+                mx = f;
+              }
+              sx += __expf(f - mx);
+            }
+          } else {
+#pragma unroll
+            for (int tj = 0; tj < 4; ++tj) {
+              const int nj = n0 + tj * 16 + li;
+              const __bf16 ob = __bf16(acc[ti][tj][r] + bv[tj]);
+              if (cv[tj]) {
+                logits[static_cast<int64_t>(mi) * V + nj] = ob;
+                const float f = static_cast<float>(ob);
+                if (f > mx) {
+                  sx *= __expf(mx - f);
+                  mx = f;
+                }
+                sx += __expf(f - mx);
+              }
+            }
+          }
+          // butterfly over the 16-lane li-group: every lane gets the
+          // row's (max, sumexp) over this 64-col stripe
+#pragma unroll
+          for (int off = 1; off < 16; off <<= 1) {
+            const float m2 = __shfl_xor(mx, off, 64);
+            const float s2 = __shfl_xor(sx, off, 64);
+            const float mn = fmaxf(mx, m2);
+            sx = (sx == 0.f ? 0.f : sx * __expf(mx - mn)) +
+                 (s2 == 0.f ? 0.f : s2 * __expf(m2 - mn));
+            mx = mn;
+          }
+          if (li == 0) {
+            float* p =
+                part + (static_cast<int64_t>(mi) * gridDim.x + blockIdx.x) *
+                           2;
+            p[0] = mx;
+            p[1] = sx;
+          }
+        }
+      }
+#pragma unroll
+      for (int ti = 0; ti < 2; ++ti)
+#pragma unroll
+        for (int tj = 0; tj < 4; ++tj) acc[ti][tj] = f32x4{0.f, 0.f, 0.f, 0.f};
+      kk = 0;
+      ++mc;
+    }
     __syncthreads();
     buf ^= 1;
-  }
-
-  // ---- epilogue: bias add, bf16 logits store, CE partial stats ----
-  // block-uniform full/tail-tile split: 238 of 239 production column
-  // tiles take the branch-free path (a per-element valid check makes
-  // hipcc branch + vmcnt(0) around every store)
-  const bool full = (n0 + mh::kBN) <= V;
-  float bv[4];
-  bool cv[4];  // column-valid (vocab tail tile)
-#pragma unroll
-  for (int tj = 0; tj < 4; ++tj) {
-    const int nj = n0 + wj * 64 + tj * 16 + li;
-    cv[tj] = nj < V;
-    bv[tj] = cv[tj] ? bias[nj] : 0.f;
-  }
-
-  // per-lane per-row (4 cols) online stats, then butterfly over the
-  // 16-lane li-group so every lane of the group holds the row stat
-  float* stats = reinterpret_cast<float*>(smem);  // [2(wi)][64 rows][2(wj)][2]
-#pragma unroll
-  for (int ti = 0; ti < 4; ++ti) {
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int mi = m0 + wi * 64 + ti * 16 + g * 4 + r;
-      float m = -INFINITY, s = 0.f;
-      if (full) {
-#pragma unroll
-        for (int tj = 0; tj < 4; ++tj) {
-          const int nj = n0 + wj * 64 + tj * 16 + li;
-          const __bf16 ob = __bf16(acc[ti][tj][r] + bv[tj]);
-          logits[static_cast<int64_t>(mi) * V + nj] = ob;
-          const float f = static_cast<float>(ob);
-          if (f > m) {
-            s *= __expf(m - f);
-            m = f;
-          }
-          s += __expf(f - m);
-        }
-      } else {
-#pragma unroll
-        for (int tj = 0; tj < 4; ++tj) {
-          const int nj = n0 + wj * 64 + tj * 16 + li;
-          const __bf16 ob = __bf16(acc[ti][tj][r] + bv[tj]);
-          if (cv[tj]) {
-            logits[static_cast<int64_t>(mi) * V + nj] = ob;
-            const float f = static_cast<float>(ob);
-            if (f > m) {
-              s *= __expf(m - f);
-              m = f;
-            }
-            s += __expf(f - m);
-          }
-        }
-      }
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1) {
-        const float m2 = __shfl_xor(m, off, 64);
-        const float s2 = __shfl_xor(s, off, 64);
-        const float mn = fmaxf(m, m2);
-        s = (s == 0.f ? 0.f : s * __expf(m - mn)) +
-            (s2 == 0.f ? 0.f : s2 * __expf(m2 - mn));
-        m = mn;
-      }
-      if (li == 0) {
-        const int row = wi * 64 + ti * 16 + g * 4 + r;
-        stats[(row * 2 + wj) * 2 + 0] = m;
-        stats[(row * 2 + wj) * 2 + 1] = s;
-      }
-    }
-  }
-  __syncthreads();
-  // fold the two wj half-tiles and publish the block's partial
-  if (tid < mh::kBM) {
-    float m1 = stats[(tid * 2 + 0) * 2 + 0], s1 = stats[(tid * 2 + 0) * 2 + 1];
-    float m2 = stats[(tid * 2 + 1) * 2 + 0], s2 = stats[(tid * 2 + 1) * 2 + 1];
-    const float mn = fmaxf(m1, m2);
-    const float sn = (s1 == 0.f ? 0.f : s1 * __expf(m1 - mn)) +
-                     (s2 == 0.f ? 0.f : s2 * __expf(m2 - mn));
-    float* p = part +
-               (static_cast<int64_t>(m0 + tid) * gridDim.y + blockIdx.y) * 2;
-    p[0] = mn;
-    p[1] = sn;
   }
 }
 
@@ -294,8 +303,7 @@ __global__ void mlm_fold_kernel(const float* __restrict__ part,
 }
 
 bool mlm_head_supported(int64_t P, int64_t V, int64_t K) {
-  return P % mh::kBM == 0 && K % mh::kBK == 0 && K >= 64 && V >= 2 &&
-         K % 8 == 0;
+  return P % mh::kChunk == 0 && K % mh::kBK == 0 && K >= 64 && V >= 2;
 }
 
 // h [P,K] bf16, w [V,K] bf16, bias [V] fp32, labels [P] int64 ->
@@ -320,7 +328,7 @@ std::vector<torch::Tensor> mlm_head_fwd(torch::Tensor h, torch::Tensor w,
   auto labels_c = labels.contiguous();
   TORCH_CHECK(labels_c.size(0) == P, "mlm_head_fwd: labels/P mismatch");
 
-  const int nTiles = (V + mh::kBN - 1) / mh::kBN;
+  const int nTiles = (V + mh::kBM - 1) / mh::kBM;
   auto fopts = h.options().dtype(torch::kFloat32);
   auto logits = torch::empty({P, V}, h.options());
   auto part = torch::empty({P, nTiles, 2}, fopts);
@@ -329,12 +337,11 @@ std::vector<torch::Tensor> mlm_head_fwd(torch::Tensor h, torch::Tensor w,
   auto lse = torch::empty({P}, fopts);
   auto stream = at::hip::getCurrentHIPStream();
 
-  const size_t lds = 4 * mh::kTile * sizeof(__bf16);
+  const size_t lds = 2 * mh::kBufB;
   HIP_CHECK(hipFuncSetAttribute(
       reinterpret_cast<const void*>(&mlm_fwd_kernel),
       hipFuncAttributeMaxDynamicSharedMemorySize, lds));
-  hipLaunchKernelGGL(mlm_fwd_kernel, dim3(P / mh::kBM, nTiles), dim3(256),
-                     lds, stream,
+  hipLaunchKernelGGL(mlm_fwd_kernel, dim3(nTiles), dim3(256), lds, stream,
                      reinterpret_cast<const __bf16*>(h.data_ptr()),
                      reinterpret_cast<const __bf16*>(w.data_ptr()),
                      bias.data_ptr<float>(),
