@@ -13,22 +13,24 @@
 // backward reuses ce_bwd (csrc/ops/cross_entropy.hip) on the bf16
 // logits this kernel stores.
 //
-// Decomposition — this shape is "M tiny, N huge": a classic 128x128
-// tiling re-streams the 62.5-MB weight matrix once per row tile
-// (P=1280 -> 10x = 625 MB of HBM, an ~80-us floor that is exactly
-// where hipBLASLt lands). Instead each block owns a 64-column vocab
-// stripe for ALL of P: grid = V/64 (~477) blocks; a block loops over P
-// in 128-row chunks with the k-loop innermost. W is then read from HBM
-// EXACTLY ONCE (62.5 MB total), its per-block 128-KB stripe stays
-// L2-resident across the row chunks, and the small h operand
-// (2.6-5 MB) is re-read by every block but from L2, not HBM. This is
-// the M=256 projection-GEMM recipe of the CDNA4 guide applied at
-// M=P: total HBM traffic ~ W + h + logits + partials ~= 150-170 MB.
+// Decomposition — this shape is "M tiny, N huge" (the guide's
+// M=256 projection-GEMM regime): each block owns a 64-column vocab
+// stripe and loops over P in 128-row chunks with the k-loop innermost,
+// so a W stripe is fetched once and re-read from cache, and the small
+// h operand (2.6-5 MB) is L2/L3-resident. Two fill levers on top:
+//   * P-split: V/64 stripes alone are ~477 blocks = 23% of the chip
+//     (PMC: waves parked 5.7x their busy cycles). The chunk loop is
+//     split nSplit ways across blocks; the same-stripe splits are
+//     placed on the SAME XCD (observed placement: XCD = block % 8,
+//     MI355X_MICROARCH "Workgroup dispatch") so the stripe's W tile is
+//     fetched into one L2 once.
+//   * 3-buffer global_load_lds pipeline with counted s_waitcnt
+//     vmcnt(N) and a raw s_barrier, keeping one 24-KB tile in flight
+//     ACROSS each barrier (a plain __syncthreads drains the DMA queue
+//     with vmcnt(0) every step).
 //
-// Staging is global_load_lds (no staging VGPRs, no ds_write pass),
-// double-buffered, one __syncthreads per 64-deep k-step. LDS images
-// are lane-linear (glds writes wave-uniform base + lane*16), so the
-// SOURCE address carries a 16-B-granule XOR swizzle:
+// LDS images are lane-linear (glds writes wave-uniform base +
+// lane*16), so the SOURCE address carries a 16-B-granule XOR swizzle:
 //   LDS[row r][granule p] = global[row r][granule p ^ (r & 7)]
 // A row is one 128-B cache line and the permutation stays inside it,
 // so coalescing is untouched; fragment b64 reads spread each 16-lane
@@ -54,12 +56,13 @@ typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4v;
 #define MFMA16(A, B, C) \
   __builtin_amdgcn_mfma_f32_16x16x32_bf16((A), (B), (C), 0, 0, 0)
 
-constexpr int kBM = 64;   // W-rows (vocab stripe) per block
+constexpr int kBN = 64;      // W-rows (vocab stripe) per block
 constexpr int kChunk = 128;  // h-rows per chunk of the in-block P loop
-constexpr int kBK = 64;   // K step (two MFMA depths per staged tile)
+constexpr int kBK = 64;      // K step (two MFMA depths per staged tile)
 constexpr int kATileB = kChunk * kBK * 2;  // h image bytes (16 KB)
-constexpr int kWTileB = kBM * kBK * 2;     // W image bytes (8 KB)
-constexpr int kBufB = kATileB + kWTileB;   // one double-buffer half
+constexpr int kWTileB = kBN * kBK * 2;     // W image bytes (8 KB)
+constexpr int kBufB = kATileB + kWTileB;   // one pipeline buffer (24 KB)
+constexpr int kGldsPerTile = 6;  // per-wave glds calls per staged tile
 
 // element byte offset of (row r, k) inside one swizzled image
 __device__ __forceinline__ int swz_off(int r, int k) {
@@ -83,31 +86,47 @@ __device__ __forceinline__ bf16x8 frag_k(const char* img, int rb, int ks) {
   return f.v;
 }
 
+#define WAIT_VM(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
+#define RAW_BARRIER()                                    \
+  do {                                                   \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");   \
+    __builtin_amdgcn_s_barrier();                        \
+  } while (0)
+
 }  // namespace mh
 
 using mh::bf16x8;
 using mh::f32x4;
 
-// grid: ceil(V/64) blocks of 256 threads (4 waves); each wave owns a
-// 32-row x 64-col sub-tile of the current 128-row chunk.
+// grid: 8 * ceil((V/64)/8) * nSplit linear blocks of 256 threads
+// (4 waves); block -> (stripe, split) keeps same-stripe splits on one
+// XCD. Each wave owns a 32-row x 64-col sub-tile of the current chunk.
 __global__ __launch_bounds__(256) void mlm_fwd_kernel(
     const __bf16* __restrict__ h,     // [P, K]
     const __bf16* __restrict__ w,     // [V, K]
     const float* __restrict__ bias,   // [V]
     __bf16* __restrict__ logits,      // [P, V]
-    float* __restrict__ part,         // [P, gridDim.x, 2] (max, sumexp)
-    int P, int V, int K) {
-  const int n0 = blockIdx.x * mh::kBM;
+    float* __restrict__ part,         // [P, nTiles, 2] (max, sumexp)
+    int P, int V, int K, int nTiles, int nSplit) {
+  const int L = blockIdx.x;
+  const int q = L >> 3;
+  const int split = q % nSplit;
+  const int stripe = (L & 7) + 8 * (q / nSplit);
+  if (stripe >= nTiles) return;  // ragged tail of the XCD mapping
+  const int n0 = stripe * mh::kBN;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
   const int g = (lane >> 4), li = lane & 15;
   const int nChunks = P / mh::kChunk;
   const int kSteps = K / mh::kBK;
+  const int myChunks = (nChunks - split + nSplit - 1) / nSplit;
+  if (myChunks <= 0) return;
+  const int total = myChunks * kSteps;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
 
   // column stripe is fixed for the whole block: bias once, in registers
-  const bool full = (n0 + mh::kBM) <= V;
+  const bool full = (n0 + mh::kBN) <= V;
   float bv[4];
   bool cv[4];
 #pragma unroll
@@ -150,23 +169,40 @@ __global__ __launch_bounds__(256) void mlm_fwd_kernel(
     }
   };
 
-  f32x4 acc[2][4] = {};
-  stage(0, 0, 0);
-  __syncthreads();
-
-  int buf = 0, kk = 0, mc = 0;
-  const int total = nChunks * kSteps;
-  for (int s = 0; s < total; ++s) {
-    // DMA the next k-slice (possibly of the next row chunk) under the
-    // MFMAs of this one
-    int kk2 = kk + 1, mc2 = mc;
-    if (kk2 == kSteps) {
-      kk2 = 0;
-      ++mc2;
+  // stage-pointer counters (chunk index is split-strided through P)
+  int st_mc = split, st_kk = 0, staged = 0;
+  auto advance_st = [&]() {
+    if (++st_kk == kSteps) {
+      st_kk = 0;
+      st_mc += nSplit;
     }
-    if (mc2 < nChunks) stage(mc2, kk2 * mh::kBK, buf ^ 1);
+    ++staged;
+  };
+  stage(st_mc, 0, 0);
+  advance_st();
+  if (staged < total) {
+    stage(st_mc, st_kk * mh::kBK, 1);
+    advance_st();
+  }
 
-    const char* at = smem + buf * mh::kBufB;
+  f32x4 acc[2][4] = {};
+  int mc = split, kk = 0;
+  for (int s = 0; s < total; ++s) {
+    // tile s must have landed; tile s+1 (6 glds) may stay in flight
+    // across the barrier — a raw barrier, NOT __syncthreads, which
+    // would emit vmcnt(0) and drain the whole DMA queue
+    if (s + 1 < total) {
+      WAIT_VM(6);
+    } else {
+      WAIT_VM(0);
+    }
+    RAW_BARRIER();
+    if (staged < total) {  // refill the buffer freed two steps ago
+      stage(st_mc, st_kk * mh::kBK, staged % 3);
+      advance_st();
+    }
+
+    const char* at = smem + (s % 3) * mh::kBufB;
     const char* wt = at + mh::kATileB;
 #pragma unroll
     for (int kd = 0; kd < 2; ++kd) {  // two MFMA depths per staged tile
@@ -236,8 +272,8 @@ __global__ __launch_bounds__(256) void mlm_fwd_kernel(
           }
           if (li == 0) {
             float* p =
-                part + (static_cast<int64_t>(mi) * gridDim.x + blockIdx.x) *
-                           2;
+                part +
+                (static_cast<int64_t>(mi) * nTiles + stripe) * 2;
             p[0] = mx;
             p[1] = sx;
           }
@@ -246,58 +282,52 @@ __global__ __launch_bounds__(256) void mlm_fwd_kernel(
 #pragma unroll
       for (int ti = 0; ti < 2; ++ti)
 #pragma unroll
-        for (int tj = 0; tj < 4; ++tj) acc[ti][tj] = f32x4{0.f, 0.f, 0.f, 0.f};
+        for (int tj = 0; tj < 4; ++tj)
+          acc[ti][tj] = f32x4{0.f, 0.f, 0.f, 0.f};
       kk = 0;
-      ++mc;
+      mc += nSplit;
     }
-    __syncthreads();
-    buf ^= 1;
   }
 }
 
-// fold per-row partials -> lse, loss (mean over non-ignored rows done
-// host-side from loss_sum / count, matching ce_fwd's contract)
+// fold per-row partials -> lse, loss: one wave per row (4 rows per
+// block), coalesced 8-B lane reads over the stripe partials.
 __global__ void mlm_fold_kernel(const float* __restrict__ part,
                                 const __bf16* __restrict__ logits,
                                 const int64_t* __restrict__ labels,
-                                float* __restrict__ loss_sum,
-                                int* __restrict__ count,
-                                float* __restrict__ lse_out, int nTiles,
-                                int V, int64_t ignore_index) {
-  const int row = blockIdx.x;
+                                float* __restrict__ sums,  // {loss, count}
+                                float* __restrict__ lse_out, int rows,
+                                int nTiles, int V, int64_t ignore_index) {
+  const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= rows) return;
+  const int lane = threadIdx.x & 63;
   const float* pr = part + static_cast<int64_t>(row) * nTiles * 2;
   float m = -INFINITY, s = 0.f;
-  for (int t = threadIdx.x; t < nTiles; t += blockDim.x) {
+  for (int t = lane; t < nTiles; t += 64) {
     const float m2 = pr[t * 2], s2 = pr[t * 2 + 1];
     const float mn = fmaxf(m, m2);
     s = (s == 0.f ? 0.f : s * __expf(m - mn)) +
         (s2 == 0.f ? 0.f : s2 * __expf(m2 - mn));
     m = mn;
   }
-  __shared__ float sm[256], ss[256];
-  sm[threadIdx.x] = m;
-  ss[threadIdx.x] = s;
-  __syncthreads();
-  for (int stride = blockDim.x / 2; stride > 0; stride >>= 1) {
-    if (threadIdx.x < stride) {
-      const float m2 = sm[threadIdx.x + stride], s2 = ss[threadIdx.x + stride];
-      const float m1 = sm[threadIdx.x], s1 = ss[threadIdx.x];
-      const float mn = fmaxf(m1, m2);
-      sm[threadIdx.x] = mn;
-      ss[threadIdx.x] = (s1 == 0.f ? 0.f : s1 * __expf(m1 - mn)) +
-                        (s2 == 0.f ? 0.f : s2 * __expf(m2 - mn));
-    }
-    __syncthreads();
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float m2 = __shfl_xor(m, off, 64);
+    const float s2 = __shfl_xor(s, off, 64);
+    const float mn = fmaxf(m, m2);
+    s = (s == 0.f ? 0.f : s * __expf(m - mn)) +
+        (s2 == 0.f ? 0.f : s2 * __expf(m2 - mn));
+    m = mn;
   }
-  if (threadIdx.x == 0) {
-    const float lse = sm[0] + __logf(ss[0]);
+  if (lane == 0) {
+    const float lse = m + __logf(s);
     lse_out[row] = lse;
     const int64_t label = labels[row];
     if (label != ignore_index) {
       const float xl = static_cast<float>(
           logits[static_cast<int64_t>(row) * V + label]);
-      atomicAdd(loss_sum, lse - xl);
-      atomicAdd(count, 1);
+      atomicAdd(&sums[0], lse - xl);
+      atomicAdd(&sums[1], 1.f);
     }
   }
 }
@@ -328,33 +358,37 @@ std::vector<torch::Tensor> mlm_head_fwd(torch::Tensor h, torch::Tensor w,
   auto labels_c = labels.contiguous();
   TORCH_CHECK(labels_c.size(0) == P, "mlm_head_fwd: labels/P mismatch");
 
-  const int nTiles = (V + mh::kBM - 1) / mh::kBM;
+  const int nTiles = (V + mh::kBN - 1) / mh::kBN;
+  const int nChunks = P / mh::kChunk;
+  // fill the chip: ~477 stripes alone are <2 blocks/CU; split the
+  // chunk loop until ~1536 blocks, bounded by the chunk count
+  int nSplit = 1;
+  while (nSplit < nChunks && nTiles * (nSplit + 1) <= 1920) ++nSplit;
+  const int grid = 8 * ((nTiles + 7) / 8) * nSplit;
+
   auto fopts = h.options().dtype(torch::kFloat32);
   auto logits = torch::empty({P, V}, h.options());
   auto part = torch::empty({P, nTiles, 2}, fopts);
-  auto loss_sum = torch::zeros({1}, fopts);
-  auto count = torch::zeros({1}, h.options().dtype(torch::kInt32));
+  auto sums = torch::zeros({2}, fopts);  // {loss_sum, count}
   auto lse = torch::empty({P}, fopts);
   auto stream = at::hip::getCurrentHIPStream();
 
-  const size_t lds = 2 * mh::kBufB;
+  const size_t lds = 3 * mh::kBufB;
   HIP_CHECK(hipFuncSetAttribute(
       reinterpret_cast<const void*>(&mlm_fwd_kernel),
       hipFuncAttributeMaxDynamicSharedMemorySize, lds));
-  hipLaunchKernelGGL(mlm_fwd_kernel, dim3(nTiles), dim3(256), lds, stream,
+  hipLaunchKernelGGL(mlm_fwd_kernel, dim3(grid), dim3(256), lds, stream,
                      reinterpret_cast<const __bf16*>(h.data_ptr()),
                      reinterpret_cast<const __bf16*>(w.data_ptr()),
                      bias.data_ptr<float>(),
                      reinterpret_cast<__bf16*>(logits.data_ptr()),
-                     part.data_ptr<float>(), P, V, K);
-  hipLaunchKernelGGL(mlm_fold_kernel, dim3(P), dim3(256), 0, stream,
-                     part.data_ptr<float>(),
+                     part.data_ptr<float>(), P, V, K, nTiles, nSplit);
+  hipLaunchKernelGGL(mlm_fold_kernel, dim3((P + 3) / 4), dim3(256), 0,
+                     stream, part.data_ptr<float>(),
                      reinterpret_cast<const __bf16*>(logits.data_ptr()),
-                     labels_c.data_ptr<int64_t>(), loss_sum.data_ptr<float>(),
-                     count.data_ptr<int>(), lse.data_ptr<float>(), nTiles, V,
-                     ignore_index);
-  return {logits, loss_sum.squeeze(0), count.squeeze(0).to(torch::kFloat32),
-          lse};
+                     labels_c.data_ptr<int64_t>(), sums.data_ptr<float>(),
+                     lse.data_ptr<float>(), P, nTiles, V, ignore_index);
+  return {logits, sums.select(0, 0), sums.select(0, 1), lse};
 }
 
 }  // namespace bpa
