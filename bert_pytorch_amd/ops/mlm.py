@@ -30,7 +30,7 @@ import torch.nn.functional as F
 from . import _reference, extension, use_native
 from .cross_entropy import fused_cross_entropy
 
-_TILE = 128  # row tile of mlm_fwd_kernel; P is padded up to it
+_TILE = 256  # row tile of mlm_fwd_kernel; P is padded up to it
 
 
 class _MlmDecoderLoss(torch.autograd.Function):
