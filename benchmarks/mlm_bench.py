@@ -38,8 +38,8 @@ def timeit(fn, iters=50, warmup=10):
 
 def main():
     ext = ops.extension()
-    # 1920 = the phase-1 bench micro-batch (local_batch 96 x 20 preds)
-    for P, V, K in [(1280, 30528, 1024), (1920, 30528, 1024),
+    # 2048 = the phase-1 bench micro-batch (1920 = 96x20 preds, kernel-padded)
+    for P, V, K in [(1280, 30528, 1024), (2048, 30528, 1024),
                     (2560, 30528, 1024)]:
         torch.manual_seed(0)
         h = (torch.randn(P, K, device=DEV) * 0.5).bfloat16()
