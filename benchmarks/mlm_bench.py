@@ -13,6 +13,8 @@ import os
 import sys
 import time
 
+os.environ["BPA_FUSED_MLM"] = "1"  # the in-repo kernel path is opt-in
+
 sys.path.insert(
     0, os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 )

@@ -21,9 +21,8 @@
 // Infinity Cache = 256 MB). The library's own pick is a 256x256
 // macro-tile, which quadruples FLOPs per staged byte; this kernel uses
 // the same: 512 threads = 8 waves as 2(M)x4(N), each wave a 128x64
-// sub-tile (acc 128 regs), BK=32 with a 3-buffer global_load_lds
-// pipeline (96 KB LDS, one block per CU) — counted s_waitcnt vmcnt(N)
-// + raw s_barrier keep one staged tile in flight ACROSS each barrier.
+// sub-tile (acc 128 regs), BK=64 double-buffered global_load_lds
+// (128 KB LDS, one block per CU), one __syncthreads per k-step.
 //
 // LDS images are lane-linear (glds writes wave-uniform base +
 // lane*16), so the SOURCE address carries a 16-B-granule XOR swizzle:
@@ -54,26 +53,20 @@ typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4v;
 
 constexpr int kBM = 256;  // h-rows per block
 constexpr int kBN = 256;  // vocab columns per block (4 x 64 stripes)
-constexpr int kBK = 32;   // K step (one MFMA depth per staged tile)
-constexpr int kATileB = kBM * kBK * 2;    // h image bytes (16 KB)
-constexpr int kWTileB = kBN * kBK * 2;    // W image bytes (16 KB)
-constexpr int kBufB = kATileB + kWTileB;  // one pipeline buffer (32 KB)
-constexpr int kGldsPerTile = 4;  // per-wave glds calls per staged tile
+constexpr int kBK = 64;   // K step (two MFMA depths per staged tile)
+constexpr int kATileB = kBM * kBK * 2;    // h image bytes (32 KB)
+constexpr int kWTileB = kBN * kBK * 2;    // W image bytes (32 KB)
+constexpr int kBufB = kATileB + kWTileB;  // one pipeline buffer (64 KB)
 
-// element byte offset of (row r, k) inside one swizzled image: a row
-// is 64 B = 4 granules, so the XOR swizzle draws on (r>>1)&3 — within
-// each row-parity class of a fragment's 16-lane group that spreads the
-// b64 reads over 4 distinct granules (2-way conflict, the minimum for
-// 64-B rows)
+// element byte offset of (row r, k) inside one swizzled image
 __device__ __forceinline__ int swz_off(int r, int k) {
-  return r * (kBK * 2) +
-         ((((k >> 3) ^ ((r >> 1) & 3)) << 4) | ((k & 7) << 1));
+  return r * (kBK * 2) + ((((k >> 3) ^ (r & 7)) << 4) | ((k & 7) << 1));
 }
 
-// fragment: lane (g = lane>>4, li = lane&15) holds row rb+li,
-// k = 4g..4g+3 and 16+4g..16+4g+3 (the probe-verified gfx950 16x16x32
-// layout).
-__device__ __forceinline__ bf16x8 frag_k(const char* img, int rb) {
+// fragment for MFMA depth base ks (0 or 32): lane (g = lane>>4,
+// li = lane&15) holds row rb+li, k = ks+4g..+3 and ks+16+4g..+3 (the
+// probe-verified gfx950 16x16x32 layout).
+__device__ __forceinline__ bf16x8 frag_k(const char* img, int rb, int ks) {
   const int lane = threadIdx.x & 63;
   const int g = (lane >> 4) & 3, li = lane & 15;
   const int r = rb + li;
@@ -81,17 +74,11 @@ __device__ __forceinline__ bf16x8 frag_k(const char* img, int rb) {
     bf16x8 v;
     bf16x4v h[2];
   } f;
-  f.h[0] = *reinterpret_cast<const bf16x4v*>(img + swz_off(r, 4 * g));
-  f.h[1] = *reinterpret_cast<const bf16x4v*>(img + swz_off(r, 16 + 4 * g));
+  f.h[0] = *reinterpret_cast<const bf16x4v*>(img + swz_off(r, ks + 4 * g));
+  f.h[1] =
+      *reinterpret_cast<const bf16x4v*>(img + swz_off(r, ks + 16 + 4 * g));
   return f.v;
 }
-
-#define WAIT_VM(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
-#define RAW_BARRIER()                                  \
-  do {                                                 \
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory"); \
-    __builtin_amdgcn_s_barrier();                      \
-  } while (0)
 
 }  // namespace mh
 
@@ -131,71 +118,61 @@ __global__ __launch_bounds__(512) void mlm_fwd_kernel(
     bv[tj] = cv[tj] ? bias[nj] : 0.f;
   }
 
-  // global_load_lds staging: per wave 2 A + 2 W calls x 16 rows x 64 B
-  const int st_sub = (lane >> 2);  // row within a 16-row piece
-  const int st_p = lane & 3;       // LDS granule within the 64-B row
+  // global_load_lds staging: per wave 4 A + 4 W calls x 8 rows x 128 B
+  const int st_sub = (lane >> 3);                // row within 8-row piece
+  const int st_swz = ((lane & 7) ^ st_sub) * 8;  // swizzled source granule
   auto stage = [&](int k0, int b) {
     char* base = smem + b * mh::kBufB;
 #pragma unroll
-    for (int c = 0; c < 2; ++c) {
-      const int r = wave * 32 + c * 16 + st_sub;  // 0..255
-      const int swz = (st_p ^ ((r >> 1) & 3)) * 8;
-      const __bf16* ga = h + static_cast<int64_t>(m0 + r) * K + k0 + swz;
+    for (int c = 0; c < 4; ++c) {
+      const int r = wave * 32 + c * 8 + st_sub;  // 0..255
+      const __bf16* ga =
+          h + static_cast<int64_t>(m0 + r) * K + k0 + st_swz;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)ga,
           (__attribute__((address_space(3))) void*)(base +
-                                                    (wave * 32 + c * 16) *
-                                                        64),
+                                                    (wave * 32 + c * 8) *
+                                                        128),
           16, 0, 0);
       // vocab tail: clamp W row (always-legal; epilogue masks)
       const __bf16* gw =
-          w + static_cast<int64_t>(min(n0 + r, V - 1)) * K + k0 + swz;
+          w + static_cast<int64_t>(min(n0 + r, V - 1)) * K + k0 + st_swz;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)gw,
           (__attribute__((address_space(3))) void*)(base + mh::kATileB +
-                                                    (wave * 32 + c * 16) *
-                                                        64),
+                                                    (wave * 32 + c * 8) *
+                                                        128),
           16, 0, 0);
     }
   };
 
-  // 3-buffer pipeline: tile s+1 (4 glds per wave) stays in flight
-  // across each raw barrier; a plain __syncthreads would emit vmcnt(0)
-  // and drain the DMA queue every step
-  const int kSteps = K / mh::kBK;
   f32x4 acc[8][4] = {};
   stage(0, 0);
-  if (kSteps > 1) stage(mh::kBK, 1);
-  int staged = kSteps > 1 ? 2 : 1;
+  __syncthreads();
 
-  for (int s = 0; s < kSteps; ++s) {
-    if (s + 1 < kSteps) {
-      WAIT_VM(4);  // tile s landed; s+1 still in flight
-    } else {
-      WAIT_VM(0);
-    }
-    RAW_BARRIER();
-    if (staged < kSteps) {  // refill the buffer freed two steps ago
-      stage(staged * mh::kBK, staged % 3);
-      ++staged;
-    }
-
-    const char* at = smem + (s % 3) * mh::kBufB;
+  int buf = 0;
+  for (int k0 = 0; k0 < K; k0 += mh::kBK) {
+    if (k0 + mh::kBK < K) stage(k0 + mh::kBK, buf ^ 1);  // DMA under MFMA
+    const char* at = smem + buf * mh::kBufB;
     const char* wt = at + mh::kATileB;
-    {
+#pragma unroll
+    for (int kd = 0; kd < 2; ++kd) {  // two MFMA depths per staged tile
       bf16x8 bfr[4];
 #pragma unroll
-      for (int t = 0; t < 4; ++t) bfr[t] = mh::frag_k(wt, wj * 64 + t * 16);
+      for (int t = 0; t < 4; ++t)
+        bfr[t] = mh::frag_k(wt, wj * 64 + t * 16, kd * 32);
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ti = 0; ti < 8; ++ti) {
-        const bf16x8 af = mh::frag_k(at, wi * 128 + ti * 16);
+        const bf16x8 af = mh::frag_k(at, wi * 128 + ti * 16, kd * 32);
 #pragma unroll
         for (int tj = 0; tj < 4; ++tj)
           acc[ti][tj] = MFMA16(af, bfr[tj], acc[ti][tj]);
       }
       __builtin_amdgcn_s_setprio(0);
     }
+    __syncthreads();
+    buf ^= 1;
   }
 
   // ---- epilogue: bias, bf16 logits, per-stripe CE stats (no LDS) ----
@@ -332,7 +309,7 @@ std::vector<torch::Tensor> mlm_head_fwd(torch::Tensor h, torch::Tensor w,
   auto lse = torch::empty({P}, fopts);
   auto stream = at::hip::getCurrentHIPStream();
 
-  const size_t lds = 3 * mh::kBufB;
+  const size_t lds = 2 * mh::kBufB;
   HIP_CHECK(hipFuncSetAttribute(
       reinterpret_cast<const void*>(&mlm_fwd_kernel),
       hipFuncAttributeMaxDynamicSharedMemorySize, lds));

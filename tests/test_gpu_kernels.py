@@ -715,7 +715,7 @@ def test_wgrad_routes_in_linear_nobias():
 )
 def test_mlm_head_fwd_kernel(P, V, K):
     torch.manual_seed(11)
-    p_pad = ((P + 127) // 128) * 128
+    p_pad = ((P + 255) // 256) * 256
     h = (torch.randn(p_pad, K, device=DEV) * 0.5).bfloat16()
     w = (torch.randn(V, K, device=DEV) * 0.05).bfloat16()
     b = torch.randn(V, device=DEV) * 0.1
@@ -741,7 +741,8 @@ def test_mlm_head_fwd_kernel(P, V, K):
 
 
 @pytest.mark.parametrize("P", [1280, 200])
-def test_mlm_decoder_loss_autograd(P):
+def test_mlm_decoder_loss_autograd(P, monkeypatch):
+    monkeypatch.setenv("BPA_FUSED_MLM", "1")  # the kernel path is opt-in
     torch.manual_seed(12)
     V, K = 30528, 1024
     h = (torch.randn(P, K, device=DEV) * 0.5).bfloat16().requires_grad_(True)
@@ -767,7 +768,8 @@ def test_mlm_decoder_loss_autograd(P):
     assert rel_err(b.grad, br.grad) < 3e-2
 
 
-def test_mlm_decoder_loss_all_ignored():
+def test_mlm_decoder_loss_all_ignored(monkeypatch):
+    monkeypatch.setenv("BPA_FUSED_MLM", "1")
     h = torch.randn(128, 256, device=DEV).bfloat16()
     w = torch.randn(512, 256, device=DEV).bfloat16()
     b = torch.zeros(512, device=DEV)
