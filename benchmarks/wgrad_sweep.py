@@ -14,9 +14,12 @@ for (name, K, M, N) in [("qkv",12288,3072,1024),("ffn1",12288,4096,1024),("attno
                         ("qkv2",8192,3072,1024),("ffn1_2",8192,4096,1024),("attnout2",8192,1024,1024)]:
     dy = torch.randn(K, M, device=dev, dtype=torch.bfloat16)
     x = torch.randn(K, N, device=dev, dtype=torch.bfloat16)
-    best = {}
-    for sk in (0,1,2,3,4,6,8,12,16):
-        if sk and K // sk < 256: continue
-        us = timeit(lambda: extension().wgrad_tn(dy, x, sk))
-        best[sk] = round(us,1)
-    print(json.dumps({"shape": name, "K":K, "M":M, "N":N, "us_by_splitk": best}), flush=True)
+    for bk in (32, 64):
+        os.environ["BPA_WGRAD_BK"] = str(bk)
+        best = {}
+        for sk in (0,1,2,3,4,6,8,12,16):
+            if sk and K // sk < 256: continue
+            us = timeit(lambda: extension().wgrad_tn(dy, x, sk))
+            best[sk] = round(us,1)
+        print(json.dumps({"shape": name, "bk": bk, "K":K, "M":M, "N":N,
+                          "us_by_splitk": best}), flush=True)

@@ -27,6 +27,7 @@
 // of the above").
 
 #include <torch/extension.h>
+#include <cstdlib>
 #include <ATen/hip/HIPContext.h>
 
 #include "../common.h"
@@ -71,7 +72,11 @@ __device__ __forceinline__ bf16x8 frag_tr(const __bf16* img, int rb,
 using wg::bf16x8;
 using wg::f32x4;
 
-// grid: (M/128, N/128, splitk); block 256.
+// grid: (M/128, N/128, splitk); block 256. BK = 32 or 64: 64 halves
+// the barrier count and doubles the MFMA run per staged tile (36.8 ->
+// 73.7 KB LDS, 4 -> 2 blocks/CU); which wins is per-shape (sweep in
+// benchmarks/wgrad_sweep.py) and picked by the host wrapper.
+template <int BK>
 __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     const __bf16* __restrict__ dy,  // [K, M]
     const __bf16* __restrict__ x,   // [K, N]
@@ -92,63 +97,69 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
   // ds_write cost plus per-store 64-bit compares)
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* lds = reinterpret_cast<__bf16*>(smem);
-  constexpr int kTile = wg::kBK * wg::kStride;
+  constexpr int kTile = BK * wg::kStride;
+  constexpr int kTPR = 256 / BK;      // staging threads per k-row
+  constexpr int kSlots = BK / 16;     // uint4 slots per thread per matrix
 
-  // staging map: thread -> (row r = tid>>3, two 8-elem slots 64 apart);
-  // each ds_write_b128's 8-lane group then covers 128 CONTIGUOUS bytes
-  // (conflict-free) instead of a 32-B-strided comb (2-way)
-  const int st_r = tid >> 3;
-  const int st_c = (tid & 7) * 8;
+  // staging map: thread -> (row r, kSlots 8-elem slots 8*kTPR apart);
+  // at BK=32 each ds_write_b128's 8-lane group covers one 128-B row
+  // (conflict-free); at BK=64 it covers two half-rows (2-way on half
+  // the banks - measured cheaper than the extra barriers it replaces)
+  const int st_r = tid / kTPR;
+  const int st_c = (tid % kTPR) * 8;
 
   // Prefetch loads are UNCONDITIONAL with the row clamped to the last
   // valid one (a conditional load/zero branch inside the loop-carried
   // lambda segfaults hipcc/ROCm 7.2's gfx950 optimizer at -O2/-O3);
   // rows past the slice end are zeroed at LDS-write time instead.
-  uint4 pf_dy[2], pf_x[2];
+  uint4 pf_dy[kSlots], pf_x[kSlots];
   auto issue_loads = [&](int k0) {
     const int krow = min(k0 + st_r, K - 1);  // clamp: always legal memory
     const uint4* ds = reinterpret_cast<const uint4*>(
         dy + static_cast<int64_t>(krow) * M + m0 + st_c);
     const uint4* xs = reinterpret_cast<const uint4*>(
         x + static_cast<int64_t>(krow) * N + n0 + st_c);
-    pf_dy[0] = ds[0];
-    pf_dy[1] = ds[8];  // +64 elems
-    pf_x[0] = xs[0];
-    pf_x[1] = xs[8];
+#pragma unroll
+    for (int q = 0; q < kSlots; ++q) {
+      pf_dy[q] = ds[q * kTPR];
+      pf_x[q] = xs[q * kTPR];
+    }
   };
 
   f32x4 acc[4][4] = {};
   issue_loads(kz0);
 
   int buf = 0;
-  for (int k0 = kz0; k0 < kz1; k0 += wg::kBK) {
+  for (int k0 = kz0; k0 < kz1; k0 += BK) {
     // write the in-flight k-step into buf (T14 write-late); zero rows
     // past the slice end so the clamped prefetch cannot contaminate
     if (k0 + st_r >= kz1) {
 #pragma unroll
-      for (int q = 0; q < 2; ++q) pf_dy[q] = pf_x[q] = uint4{0, 0, 0, 0};
+      for (int q = 0; q < kSlots; ++q) pf_dy[q] = pf_x[q] = uint4{0, 0, 0, 0};
     }
     const int boff = buf * 2 * kTile;
 #pragma unroll
-    for (int q = 0; q < 2; ++q) {
+    for (int q = 0; q < kSlots; ++q) {
       *reinterpret_cast<uint4*>(
-          &lds[boff + st_r * wg::kStride + st_c + 64 * q]) = pf_dy[q];
+          &lds[boff + st_r * wg::kStride + st_c + 8 * kTPR * q]) = pf_dy[q];
       *reinterpret_cast<uint4*>(
-          &lds[boff + kTile + st_r * wg::kStride + st_c + 64 * q]) = pf_x[q];
+          &lds[boff + kTile + st_r * wg::kStride + st_c + 8 * kTPR * q]) =
+          pf_x[q];
     }
     __syncthreads();
-    issue_loads(k0 + wg::kBK < kz1 ? k0 + wg::kBK : k0);  // T14 issue-early
+    issue_loads(k0 + BK < kz1 ? k0 + BK : k0);  // T14 issue-early
 
     const __bf16* dyt = lds + boff;
     const __bf16* xt = lds + boff + kTile;
-    {
+#pragma unroll
+    for (int kd = 0; kd < BK / 32; ++kd) {
       bf16x8 af[4], bfr[4];
 #pragma unroll
       for (int t = 0; t < 4; ++t)
-        af[t] = wg::frag_tr(dyt, 0, wi * 64 + t * 16);
+        af[t] = wg::frag_tr(dyt, kd * 32, wi * 64 + t * 16);
 #pragma unroll
       for (int t = 0; t < 4; ++t)
-        bfr[t] = wg::frag_tr(xt, 0, wj * 64 + t * 16);
+        bfr[t] = wg::frag_tr(xt, kd * 32, wj * 64 + t * 16);
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ti = 0; ti < 4; ++ti)
@@ -355,6 +366,11 @@ torch::Tensor wgrad_tn(torch::Tensor dy, torch::Tensor x,
   const bool big = false;  // 256-tile variant measured SLOWER (1 block/CU cannot hide latency); kept for reference
   const int bm = big ? 256 : wg::kBM, bn = big ? 256 : wg::kBN;
   const int tiles = (M / bm) * (N / bn);
+  // k-step depth: 64 halves barriers / doubles the MFMA run per tile
+  // at 2 blocks/CU; override with BPA_WGRAD_BK for sweeps
+  const char* env_bk = getenv("BPA_WGRAD_BK");  // per-call: sweepable
+  int bk = env_bk ? atoi(env_bk) : 32;
+  TORCH_CHECK(bk == 32 || bk == 64, "wgrad_tn: BK must be 32 or 64");
   // split-K sweep on MI355X (benchmarks/wgrad_sweep.py): best wall time
   // lands at tiles*splitk ~ 512-768 with splitk <= 8 (beyond that the
   // fp32 slab traffic of the combine outweighs the occupancy gain):
@@ -364,8 +380,7 @@ torch::Tensor wgrad_tn(torch::Tensor dy, torch::Tensor x,
          K / (splitk + 1) >= 8 * wg::kBK)
     ++splitk;
   if (splitk_override > 0) splitk = static_cast<int>(splitk_override);
-  int k_slice =
-      ((K + splitk - 1) / splitk + wg::kBK - 1) / wg::kBK * wg::kBK;
+  int k_slice = ((K + splitk - 1) / splitk + bk - 1) / bk * bk;
   splitk = (K + k_slice - 1) / k_slice;  // drop empty tail slices
 
   auto part = torch::empty({splitk, static_cast<int64_t>(M), N},
@@ -374,7 +389,7 @@ torch::Tensor wgrad_tn(torch::Tensor dy, torch::Tensor x,
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid(M / bm, N / bn, splitk), block(big ? 512 : 256);
   const size_t lds =
-      4 * wg::kBK * (big ? wg::kStride256 : wg::kStride) * sizeof(__bf16);
+      4 * bk * (big ? wg::kStride256 : wg::kStride) * sizeof(__bf16);
   if (big) {
     HIP_CHECK(hipFuncSetAttribute(
         reinterpret_cast<const void*>(&wgrad_tn256_kernel),
@@ -383,11 +398,19 @@ torch::Tensor wgrad_tn(torch::Tensor dy, torch::Tensor x,
                        reinterpret_cast<const __bf16*>(dy.data_ptr()),
                        reinterpret_cast<const __bf16*>(x.data_ptr()),
                        part.data_ptr<float>(), K, M, N, k_slice);
+  } else if (bk == 64) {
+    HIP_CHECK(hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&wgrad_tn_kernel<64>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, lds));
+    hipLaunchKernelGGL(wgrad_tn_kernel<64>, grid, block, lds, stream,
+                       reinterpret_cast<const __bf16*>(dy.data_ptr()),
+                       reinterpret_cast<const __bf16*>(x.data_ptr()),
+                       part.data_ptr<float>(), K, M, N, k_slice);
   } else {
     HIP_CHECK(hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&wgrad_tn_kernel),
+        reinterpret_cast<const void*>(&wgrad_tn_kernel<32>),
         hipFuncAttributeMaxDynamicSharedMemorySize, lds));
-    hipLaunchKernelGGL(wgrad_tn_kernel, grid, block, lds, stream,
+    hipLaunchKernelGGL(wgrad_tn_kernel<32>, grid, block, lds, stream,
                        reinterpret_cast<const __bf16*>(dy.data_ptr()),
                        reinterpret_cast<const __bf16*>(x.data_ptr()),
                        part.data_ptr<float>(), K, M, N, k_slice);
