@@ -196,3 +196,95 @@ def test_grad_compress_hook_gloo():
         p.join(timeout=30)
     for rank, ok in results:
         assert ok is True, f"rank {rank}: {ok}"
+
+
+def _run_ddp_fused_loss_step(rank, world, port, tmpdir, q):
+    """DDP grad-sync equivalence through the compute_mlm_loss=True path
+    (the runner/bench path: model returns the scalar MLM loss and the
+    criterion's 0-dim branch adds only NSP)."""
+    try:
+        _init(rank, world, port)
+        torch.manual_seed(0)
+        from bert_pytorch_amd.config import BertConfig
+        from bert_pytorch_amd.models import (
+            BertForPreTraining,
+            BertPretrainingCriterion,
+        )
+
+        config = BertConfig(
+            vocab_size_or_config_json_file=256, hidden_size=32,
+            num_hidden_layers=2, num_attention_heads=2,
+            intermediate_size=64, max_position_embeddings=32,
+        )
+        model = BertForPreTraining(config)
+        ddp = torch.nn.parallel.DistributedDataParallel(model)
+        criterion = BertPretrainingCriterion(config.vocab_size)
+
+        g = torch.Generator().manual_seed(100 + rank)
+        ids = torch.randint(0, 256, (2, 16), generator=g)
+        mask = torch.ones(2, 16, dtype=torch.long)
+        labels = torch.full((2, 16), -1, dtype=torch.long)
+        labels[:, 3] = 5
+        nsp = torch.zeros(2, dtype=torch.long)
+
+        model.eval()
+        mlm_loss, rel, glabels = ddp(
+            ids, None, mask, masked_lm_labels=labels,
+            max_predictions_per_seq=4, compute_mlm_loss=True,
+        )
+        assert mlm_loss.dim() == 0
+        loss = criterion(mlm_loss, rel, glabels, nsp)
+        loss.backward()
+
+        torch.manual_seed(0)
+        solo = BertForPreTraining(config)
+        solo.load_state_dict(model.state_dict())
+        solo.eval()
+        s2, r2, l2 = solo(
+            ids, None, mask, masked_lm_labels=labels,
+            max_predictions_per_seq=4, compute_mlm_loss=True,
+        )
+        criterion(s2, r2, l2, nsp).backward()
+        # check the tied embedding/decoder weight AND the MLM bias —
+        # both only get grads through the fused-loss path
+        checked = 0
+        ok = True
+        for name, p_ddp in ddp.module.named_parameters():
+            if name not in (
+                "bert.embeddings.word_embeddings.weight",
+                "cls.predictions.bias",
+            ):
+                continue
+            p_solo = dict(solo.named_parameters())[name]
+            manual = p_solo.grad.clone()
+            dist.all_reduce(manual)
+            manual /= world
+            ok = ok and torch.allclose(
+                p_ddp.grad, manual, rtol=1e-5, atol=1e-6
+            )
+            checked += 1
+        q.put((rank, bool(ok and checked == 2), float(loss)))
+        dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"ERROR: {e}", 0.0))
+
+
+@pytest.mark.timeout(120)
+def test_ddp_fused_mlm_loss_grad_sync_gloo(tmp_path):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29517
+    procs = [
+        ctx.Process(
+            target=_run_ddp_fused_loss_step,
+            args=(r, WORLD, port, str(tmp_path), q),
+        )
+        for r in range(WORLD)
+    ]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=110) for _ in range(WORLD)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, ok, loss in results:
+        assert ok is True, f"rank {rank}: {ok}"
