@@ -673,7 +673,9 @@ def test_attention_dropout_numerics(B, S, NH):
     (4100, 1024, 1024),    # K not a multiple of 64: zero-pad tail path
     (384, 128, 128),       # minimal tile
 ])
-def test_wgrad_tn_parity(K, M, N):
+@pytest.mark.parametrize("bk", [32, 64])
+def test_wgrad_tn_parity(K, M, N, bk, monkeypatch):
+    monkeypatch.setenv("BPA_WGRAD_BK", str(bk))
     torch.manual_seed(K + M + N)
     dy = torch.randn(K, M, device=DEV, dtype=torch.bfloat16)
     x = torch.randn(K, N, device=DEV, dtype=torch.bfloat16)
