@@ -238,7 +238,7 @@ __global__ __launch_bounds__(512) void mlm_fwd_kernel(
 __global__ void mlm_fold_kernel(const float* __restrict__ part,
                                 const __bf16* __restrict__ logits,
                                 const int64_t* __restrict__ labels,
-                                float* __restrict__ sums,  // {loss, count}
+                                float* __restrict__ row_out,  // [2, rows]
                                 float* __restrict__ lse_out, int rows,
                                 int nStripes, int V, int64_t ignore_index) {
   const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
@@ -266,12 +266,40 @@ __global__ void mlm_fold_kernel(const float* __restrict__ part,
     const float lse = m + __logf(s);
     lse_out[row] = lse;
     const int64_t label = labels[row];
-    if (label != ignore_index) {
-      const float xl = static_cast<float>(
-          logits[static_cast<int64_t>(row) * V + label]);
-      atomicAdd(&sums[0], lse - xl);
-      atomicAdd(&sums[1], 1.f);
+    // per-row (loss, valid) instead of same-address atomics (2 x rows
+    // serialized L2 atomics); mlm_sum_kernel folds them
+    const bool valid = label != ignore_index;
+    const float xl = valid ? static_cast<float>(
+                                 logits[static_cast<int64_t>(row) * V +
+                                        (valid ? label : 0)])
+                           : 0.f;
+    row_out[row] = valid ? lse - xl : 0.f;
+    row_out[rows + row] = valid ? 1.f : 0.f;
+  }
+}
+
+// fold [2, rows] (loss, valid) -> sums [2] = {loss_sum, count}
+__global__ void mlm_sum_kernel(const float* __restrict__ row_out,
+                               float* __restrict__ sums, int rows) {
+  float l = 0.f, c = 0.f;
+  for (int i = threadIdx.x; i < rows; i += blockDim.x) {
+    l += row_out[i];
+    c += row_out[rows + i];
+  }
+  __shared__ float sl[256], sc[256];
+  sl[threadIdx.x] = l;
+  sc[threadIdx.x] = c;
+  __syncthreads();
+  for (int stride = blockDim.x / 2; stride > 0; stride >>= 1) {
+    if (threadIdx.x < stride) {
+      sl[threadIdx.x] += sl[threadIdx.x + stride];
+      sc[threadIdx.x] += sc[threadIdx.x + stride];
     }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    sums[0] = sl[0];
+    sums[1] = sc[0];
   }
 }
 
@@ -305,7 +333,8 @@ std::vector<torch::Tensor> mlm_head_fwd(torch::Tensor h, torch::Tensor w,
   auto fopts = h.options().dtype(torch::kFloat32);
   auto logits = torch::empty({P, V}, h.options());
   auto part = torch::empty({P, nStripes, 2}, fopts);
-  auto sums = torch::zeros({2}, fopts);  // {loss_sum, count}
+  auto row_out = torch::empty({2, P}, fopts);
+  auto sums = torch::empty({2}, fopts);  // {loss_sum, count}
   auto lse = torch::empty({P}, fopts);
   auto stream = at::hip::getCurrentHIPStream();
 
@@ -324,8 +353,10 @@ std::vector<torch::Tensor> mlm_head_fwd(torch::Tensor h, torch::Tensor w,
   hipLaunchKernelGGL(mlm_fold_kernel, dim3((P + 3) / 4), dim3(256), 0,
                      stream, part.data_ptr<float>(),
                      reinterpret_cast<const __bf16*>(logits.data_ptr()),
-                     labels_c.data_ptr<int64_t>(), sums.data_ptr<float>(),
+                     labels_c.data_ptr<int64_t>(), row_out.data_ptr<float>(),
                      lse.data_ptr<float>(), P, nStripes, V, ignore_index);
+  hipLaunchKernelGGL(mlm_sum_kernel, dim3(1), dim3(256), 0, stream,
+                     row_out.data_ptr<float>(), sums.data_ptr<float>(), P);
   return {logits, sums.select(0, 0), sums.select(0, 1), lse};
 }
 
