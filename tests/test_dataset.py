@@ -166,3 +166,26 @@ def test_legacy_premasked_format(tmp_path):
     ids, seg, mask, labels, nsp = ds[0]
     assert labels[2] == 7 and labels[5] == 9
     assert (labels != -1).sum() == 2
+
+
+def test_corrupt_and_missing_shards_are_skipped(shards, tmp_path):
+    """Failure handling: unreadable / truncated / missing
+    shards are skipped with a warning; valid ones still serve."""
+    bad_garbage = tmp_path / "garbage.hdf5"
+    bad_garbage.write_bytes(b"\x00\x01not an hdf5 file\xff" * 64)
+    bad_trunc = tmp_path / "truncated.hdf5"
+    bad_trunc.write_bytes(open(shards[0], "rb").read()[:200])
+    missing = str(tmp_path / "never_written.hdf5")
+    files = [shards[0], str(bad_garbage), str(bad_trunc), missing, shards[1]]
+    with pytest.warns(UserWarning):
+        ds = _dataset(files)
+    assert len(ds) == 32  # the two valid shards
+    ids, seg, mask, labels, nsp = ds[0]
+    assert ids.shape[0] == 32
+
+
+def test_all_shards_invalid_raises(tmp_path):
+    bad = tmp_path / "junk.hdf5"
+    bad.write_bytes(b"junk")
+    with pytest.raises(RuntimeError), pytest.warns(UserWarning):
+        _dataset([str(bad)])
