@@ -777,3 +777,43 @@ def test_mlm_decoder_loss_all_ignored(monkeypatch):
     b = torch.zeros(512, device=DEV)
     labels = torch.full((128,), -1, device=DEV, dtype=torch.long)
     assert float(ops.mlm_decoder_loss(h, w, b, labels)) == 0.0
+
+
+# ---------------------------------------------------------------------------
+# RCCL on-device smoke (the "nccl" backend on ROCm IS RCCL)
+# ---------------------------------------------------------------------------
+def test_rccl_collectives_on_device(monkeypatch):
+    """RCCL init + all_reduce/broadcast/all_gather + a DDP-wrapped
+    forward/backward on the MI355X. The pool leases single-GPU boxes
+    (RCCL does not allow two ranks on one device), so this is world=1:
+    it verifies the RCCL library loads, a communicator forms, and the
+    exact collective calls the multi-GPU path issues execute on-device;
+    multi-rank semantics are covered by the gloo world-2/4/8 suites
+    (tests/test_distributed_cpu.py, test_distributed_runner.py) and the
+    driver's round-end SCALE run."""
+    import torch.distributed as dist
+
+    monkeypatch.setenv("MASTER_ADDR", "127.0.0.1")
+    monkeypatch.setenv("MASTER_PORT", "29531")
+    monkeypatch.setenv("RANK", "0")
+    monkeypatch.setenv("WORLD_SIZE", "1")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        x = torch.arange(1024, device=DEV, dtype=torch.float32)
+        dist.all_reduce(x)
+        assert torch.equal(x, torch.arange(1024, device=DEV).float())
+        dist.broadcast(x, src=0)
+        out = [torch.empty_like(x)]
+        dist.all_gather(out, x)
+        assert torch.equal(out[0], x)
+
+        m = torch.nn.Linear(64, 64).to(DEV)
+        ddp = torch.nn.parallel.DistributedDataParallel(
+            m, device_ids=[0], gradient_as_bucket_view=True
+        )
+        y = ddp(torch.randn(8, 64, device=DEV)).sum()
+        y.backward()
+        assert m.weight.grad is not None
+        assert torch.isfinite(m.weight.grad).all()
+    finally:
+        dist.destroy_process_group()
