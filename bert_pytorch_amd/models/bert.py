@@ -478,6 +478,14 @@ class BertForPreTraining(BertPreTrainedModel):
     [B,S,V] head — unmasked rows contribute zero — at ~1/6 of the
     decoder-GEMM FLOPs for phase-1 shapes) and the gathered labels are
     returned alongside the scores.
+
+    With ``compute_mlm_loss=True`` (the runner/bench path) the first
+    return element is the scalar MLM loss instead of the [P, V] scores:
+    the decoder GEMM + bias + cross-entropy run through
+    ``ops.mlm_decoder_loss`` (library GEMM + fused-CE kernel by
+    default; the in-repo fused MFMA kernel with BPA_FUSED_MLM=1), and
+    ``BertPretrainingCriterion`` detects the 0-dim tensor and adds only
+    the NSP term.
     """
 
     def __init__(self, config: BertConfig):
