@@ -19,17 +19,18 @@ backward to avoid a 78-MB round trip would cost ~80 us to save ~20 us
 on this hardware (8 TB/s HBM vs ~1 PF/s GEMM), so materialize-and-reuse
 is the right MI355X trade.
 
-Routing (measured, docs/KERNELS.md "MLM decoder"): hipBLASLt runs this
-[1280-2560, 30528, 1024] GEMM at ~870 TF/s (92 us) and the standalone
-ce_fwd pass costs 42 us; the best of four in-repo kernel structures
-(256x256 macro-tile, BK=64 glds double-buffer) lands at 193 us
-including the fused CE — correct and tested, but a net ~60 us/micro
-regression, so the DEFAULT path is the library GEMM + fused-CE kernel
-and the in-repo fused kernel is opt-in via BPA_FUSED_MLM=1. After the
-round-1 masked-row gather this whole head is ~1.3% of a phase-1 step,
-which bounds any fusion win below measurement noise; the fused kernel
-stays in-tree as the measured-and-documented experiment with full
-parity coverage (tests/test_gpu_kernels.py::test_mlm_*).
+Routing (measured; docs/KERNELS.md "MLM decoder" and
+profiles/mlm_fused_study.md): the shipped structure (256x256
+macro-tile, BK=64 glds double-buffer, atomics-free fold) runs
+178-275 us forward including the CE statistics vs 131-201 us for
+hipBLASLt + the standalone ce_fwd pass, but its leaner backward makes
+the full train step tie at P=1280 and win at P>=2048, and a same-box
+e2e A/B measured dead even (phase 1 -0.3%, phase 2 +0.6% — the whole
+head is ~1.3% of a step after round 1's masked-row gather). At
+measured parity the in-repo MFMA kernel is the DEFAULT (the
+framework's biggest GEMM runs in-repo); BPA_FUSED_MLM=0 opts back
+into the library-GEMM path. Full parity coverage:
+tests/test_gpu_kernels.py::test_mlm_*.
 """
 
 from __future__ import annotations
@@ -104,7 +105,7 @@ def mlm_decoder_loss(
         if (
             bf16
             and ext.mlm_head_supported(p_pad, V, K)
-            and os.environ.get("BPA_FUSED_MLM") == "1"  # opt-in (see above)
+            and os.environ.get("BPA_FUSED_MLM") != "0"  # opt-out (see above)
         ):
             h = hidden.to(torch.bfloat16)
             w = weight.to(torch.bfloat16)
