@@ -111,8 +111,17 @@ def mlm_decoder_loss(
             w = weight.to(torch.bfloat16)
             b = bias.float()
             return _MlmDecoderLoss.apply(h, w, b, labels, ignore_index)
-        # unfused native: library GEMM + fused CE kernel
-        scores = F.linear(hidden, weight, bias)
+        # unfused native: library GEMM + fused CE kernel. Outside
+        # autocast F.linear requires matching dtypes, so align the fp32
+        # bias with bf16 inputs explicitly (autograd casts the grad
+        # back to fp32).
+        w, b = weight, bias
+        if not torch.is_autocast_enabled():
+            if w.dtype != hidden.dtype:
+                w = w.to(hidden.dtype)
+            if b.dtype != hidden.dtype:
+                b = b.to(hidden.dtype)
+        scores = F.linear(hidden, w, b)
         return fused_cross_entropy(scores, labels, ignore_index)
     # CPU / eager oracle
     scores = F.linear(hidden.float(), weight.float(), bias.float())
