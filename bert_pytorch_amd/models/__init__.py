@@ -16,3 +16,4 @@ from .bert import (  # noqa: F401
     LinearActivation,
 )
 from .criterion import BertPretrainingCriterion  # noqa: F401
+from .tf_import import load_tf_weights, tf_name_to_state_key  # noqa: F401
