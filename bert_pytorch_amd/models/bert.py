@@ -410,10 +410,20 @@ class BertPreTrainedModel(nn.Module):
                 module._checkpoint_activations = flag
 
     @classmethod
-    def from_pretrained(cls, path: str, config: Optional[BertConfig] = None, **kw):
+    def from_pretrained(
+        cls,
+        path: str,
+        config: Optional[BertConfig] = None,
+        from_tf: bool = False,
+        **kw,
+    ):
         """Load from a local checkpoint: a ``ckpt_*.pt`` dict with a
-        'model' key, or a bare state-dict file. (The reference's
-        URL/S3/TF import paths need network access, absent here.)"""
+        'model' key, a bare state-dict file, or — with ``from_tf=True``
+        or when ``path`` points at a TF bundle prefix (``*.ckpt`` with
+        an ``.index`` next to it) — a TensorFlow checkpoint parsed by
+        the in-repo bundle reader (reference: src/modeling.py:58-116,
+        no TensorFlow needed). The reference's URL/S3 download cache
+        needs network access, absent here."""
         import os
 
         if config is None:
@@ -423,6 +433,10 @@ class BertPreTrainedModel(nn.Module):
             else:
                 raise ValueError("config required when path is not a model dir")
         model = cls(config, **kw)
+        if from_tf or os.path.isfile(path + ".index"):
+            from .tf_import import load_tf_weights  # noqa: PLC0415
+
+            return load_tf_weights(model, path, strict=False)
         sd_file = path
         if os.path.isdir(path):
             sd_file = os.path.join(path, "pytorch_model.bin")

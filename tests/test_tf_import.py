@@ -158,3 +158,17 @@ def test_load_tf_weights_end_to_end(tiny_config, tmp_path):
         s2, r2 = dst(ids, None, mask)
     torch.testing.assert_close(s1, s2)
     torch.testing.assert_close(r1, r2)
+
+
+def test_from_pretrained_tf_prefix(tiny_config, tmp_path):
+    """from_pretrained auto-detects a TF bundle prefix (reference's
+    from_tf path) and loads it through the in-repo parser."""
+    torch.manual_seed(2)
+    src = BertForPreTraining(tiny_config).eval()
+    prefix = str(tmp_path / "bert_model.ckpt")
+    _export_tf_style(src, prefix)
+    dst = BertForPreTraining.from_pretrained(prefix, config=tiny_config)
+    torch.testing.assert_close(
+        dst.state_dict()["bert.embeddings.word_embeddings.weight"],
+        src.state_dict()["bert.embeddings.word_embeddings.weight"],
+    )
