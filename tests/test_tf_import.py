@@ -172,3 +172,25 @@ def test_from_pretrained_tf_prefix(tiny_config, tmp_path):
         dst.state_dict()["bert.embeddings.word_embeddings.weight"],
         src.state_dict()["bert.embeddings.word_embeddings.weight"],
     )
+
+
+def test_bundle_malformed_files(tmp_path):
+    short = tmp_path / "short.ckpt.index"
+    short.write_bytes(b"tiny")
+    with pytest.raises(ValueError, match="too short"):
+        TFBundleReader(str(tmp_path / "short.ckpt"))
+    bad = tmp_path / "bad.ckpt.index"
+    bad.write_bytes(b"\x00" * 64)
+    with pytest.raises(ValueError, match="magic"):
+        TFBundleReader(str(tmp_path / "bad.ckpt"))
+
+
+def test_snappy_copy2_copy4():
+    # literal "ab" then copy2(offset 2, len 6) -> "abababab"
+    comp = bytes([8, 0b00000100]) + b"ab" + \
+        bytes([0b00010110]) + (2).to_bytes(2, "little")
+    assert snappy_decompress(comp) == b"abababab"
+    # literal "xy" then copy4(offset 2, len 2) -> "xyxy"
+    comp = bytes([4, 0b00000100]) + b"xy" + \
+        bytes([0b00000111]) + (2).to_bytes(4, "little")
+    assert snappy_decompress(comp) == b"xyxy"
