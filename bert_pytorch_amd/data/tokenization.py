@@ -135,7 +135,7 @@ class WordpieceTokenizer:
     fallback; reference: src/tokenization.py:176-229)."""
 
     def __init__(self, vocab: Dict[str, int], unk_token: str = "[UNK]",
-                 max_input_chars_per_word: int = 200):
+                 max_input_chars_per_word: int = 100):
         self.vocab = vocab
         self.unk_token = unk_token
         self.max_input_chars_per_word = max_input_chars_per_word

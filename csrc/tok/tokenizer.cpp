@@ -25,7 +25,7 @@ namespace bpa_tok {
 struct WordPieceModel {
   std::unordered_map<std::string, int64_t> vocab;
   std::string unk = "[UNK]";
-  int max_input_chars = 200;
+  int max_input_chars = 100;  // reference: src/tokenization.py:179
 };
 
 struct BPEModel {
