@@ -301,22 +301,35 @@ class ByteLevelBPETokenizer:
     def id_to_token(self, idx: int) -> Optional[str]:
         return self.ids_to_tokens.get(idx)
 
+    # GPT-2 / RoBERTa pre-tokenization pattern (contractions, letter
+    # runs, digit runs, punctuation runs — each optionally taking one
+    # leading space). Needs \p classes, so the `regex` module; without
+    # it a simplified whitespace-prefix split is used (documented
+    # divergence: punctuation is then folded into words).
+    _GPT2_PAT = (
+        r"""'s|'t|'re|'ve|'m|'ll|'d|"""
+        r""" ?\p{L}+| ?\p{N}+| ?[^\s\p{L}\p{N}]+|\s+(?!\S)|\s+"""
+    )
+
     def _pretokenize(self, text: str) -> List[str]:
         if self.lowercase:
             text = text.lower()
-        # whitespace-prefix pretokenization (space folded into the token,
-        # GPT-2 style) then byte-mapping
-        words: List[str] = []
-        current = ""
-        for ch in text:
-            if ch == " ":
-                if current:
-                    words.append(current)
-                current = " "
-            else:
-                current += ch
-        if current:
-            words.append(current)
+        try:
+            import regex  # noqa: PLC0415
+
+            words = regex.findall(self._GPT2_PAT, text)
+        except ImportError:  # pragma: no cover - regex ships in-image
+            words = []
+            current = ""
+            for ch in text:
+                if ch == " ":
+                    if current:
+                        words.append(current)
+                    current = " "
+                else:
+                    current += ch
+            if current:
+                words.append(current)
         return [
             "".join(_BYTE_ENC[b] for b in w.encode("utf-8")) for w in words
         ]

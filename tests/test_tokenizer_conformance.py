@@ -88,3 +88,78 @@ def test_ids_match_hf_rust(vocab_file):
     for text in TEXTS:
         enc = hf.encode(text, add_special_tokens=False)
         assert ours.convert_tokens_to_ids(ours.tokenize(text)) == enc.ids, text
+
+
+def test_cpp_wordpiece_class_matches_hf_rust(vocab_file):
+    """The C++-backed WordPieceTokenizer (the pretraining encode path)
+    agrees with the Rust pipeline too."""
+    from bert_pytorch_amd.data.tokenization import WordPieceTokenizer
+
+    ours = WordPieceTokenizer(vocab_file, lowercase=True)
+    hf = _hf_bert_pipeline(vocab_file, True)
+    for text in TEXTS:
+        got = ours.tokenize(text)
+        want = hf.encode(text, add_special_tokens=False).tokens
+        assert got == want, (text, got, want)
+
+
+def test_byte_bpe_matches_hf_rust(tmp_path):
+    """RoBERTa-style byte-level BPE: train vocab+merges with the Rust
+    library, encode with both it and the in-repo tokenizer (GPT-2
+    regex pre-tokenization + C++ merge core)."""
+    from tokenizers import Tokenizer
+    from tokenizers.models import BPE
+    from tokenizers.pre_tokenizers import ByteLevel
+    from tokenizers.decoders import ByteLevel as ByteLevelDecoder
+    from tokenizers.trainers import BpeTrainer
+
+    from bert_pytorch_amd.data.tokenization import ByteLevelBPETokenizer
+
+    corpus = [
+        "the quick brown fox jumps over the lazy dog",
+        "it's John's book, isn't it? Yes -- really!",
+        "numbers 1999 and 42 mix with words2000 tokens",
+        "Hello, world! hello WORLD... the theater there",
+    ] * 8
+    corpus_file = tmp_path / "corpus.txt"
+    corpus_file.write_text("\n".join(corpus), encoding="utf-8")
+
+    hf = Tokenizer(BPE(unk_token=None))
+    hf.pre_tokenizer = ByteLevel(add_prefix_space=False)
+    hf.decoder = ByteLevelDecoder()
+    trainer = BpeTrainer(
+        vocab_size=400, min_frequency=2, special_tokens=["<s>", "</s>"],
+        initial_alphabet=ByteLevel.alphabet(),
+    )
+    hf.train([str(corpus_file)], trainer)
+
+    import json
+
+    model_dir = tmp_path / "bpe"
+    model_dir.mkdir()
+    blob = json.loads(hf.to_str())
+    (model_dir / "vocab.json").write_text(
+        json.dumps(blob["model"]["vocab"]), encoding="utf-8"
+    )
+    merge_lines = [
+        m if isinstance(m, str) else " ".join(m)
+        for m in blob["model"]["merges"]
+    ]
+    (model_dir / "merges.txt").write_text(
+        "\n".join(merge_lines) + "\n", encoding="utf-8"
+    )
+
+    ours = ByteLevelBPETokenizer(
+        str(model_dir / "vocab.json"), str(model_dir / "merges.txt")
+    )
+    texts = corpus[:4] + [
+        "unseen words zigzag quickly!",
+        "  leading spaces and   runs",
+        "punct,comma.dot-dash's",
+        "",
+    ]
+    for text in texts:
+        want = hf.encode(text).tokens
+        got = ours.tokenize(text)
+        assert got == want, (text, got, want)
+        assert [ours.token_to_id(t) for t in got] == hf.encode(text).ids
