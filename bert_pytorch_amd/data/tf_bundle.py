@@ -242,6 +242,8 @@ class TFBundleReader:
 
     @staticmethod
     def _block(raw: bytes, off: int, size: int) -> bytes:
+        if off + size + 5 > len(raw):  # contents + 5-byte trailer
+            raise ValueError("truncated SSTable block")
         contents = raw[off:off + size]
         ctype = raw[off + size]  # 1-byte type in the 5-byte trailer
         if ctype == 1:
@@ -290,6 +292,8 @@ class TFBundleReader:
         with open(shard_path, "rb") as f:
             f.seek(e["offset"])
             buf = f.read(e["size"])
+        if len(buf) != e["size"]:
+            raise ValueError(f"{shard_path}: truncated at {name!r}")
         arr = np.frombuffer(buf, dtype=dt)
         return arr.reshape(e["shape"]) if e["shape"] else arr.reshape(())
 

@@ -194,3 +194,16 @@ def test_snappy_copy2_copy4():
     comp = bytes([4, 0b00000100]) + b"xy" + \
         bytes([0b00000111]) + (2).to_bytes(4, "little")
     assert snappy_decompress(comp) == b"xyxy"
+
+
+def test_bundle_truncated_data_shard(tmp_path):
+    prefix = str(tmp_path / "m.ckpt")
+    w = TFBundleWriter(prefix)
+    w.add("t", np.arange(64, dtype=np.float32))
+    w.save()
+    data = prefix + ".data-00000-of-00001"
+    with open(data, "r+b") as f:
+        f.truncate(16)
+    r = TFBundleReader(prefix)
+    with pytest.raises(ValueError, match="truncated"):
+        r.load_variable("t")
